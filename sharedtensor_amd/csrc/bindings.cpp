@@ -1,0 +1,119 @@
+// Python bindings for the shared-tensor engine (replaces the reference's Lua
+// C-API layer, /root/reference/src/sharedtensor.c:347-477).  The binding is
+// torch-header-free: tensors cross the boundary as raw data_ptr() integers,
+// so the extension builds with bare hipcc and works for CPU and GPU tensors
+// alike.  All blocking entrypoints release the GIL; engine threads never
+// touch Python.
+#include <pybind11/pybind11.h>
+#include <pybind11/stl.h>
+
+#include "codec_cpu.h"
+#include "engine.h"
+
+namespace py = pybind11;
+using namespace shamd;
+
+namespace {
+
+// CPU codec entrypoints for tests (numerics parity vs ops/oracle.py).
+py::tuple py_cpu_encode(int codec, uintptr_t delta, int64_t n, double scale_in,
+                        uintptr_t payload_out) {
+  Codec c = static_cast<Codec>(codec);
+  float* d = reinterpret_cast<float*>(delta);
+  float scale = scale_in < 0 ? cpu_compute_scale(c, d, n)
+                             : static_cast<float>(scale_in);
+  cpu_quantize(c, d, n, scale, reinterpret_cast<uint8_t*>(payload_out));
+  return py::make_tuple(scale, payload_bytes(c, n));
+}
+
+void py_cpu_apply(int codec, uintptr_t payload, int64_t n, double scale,
+                  std::vector<uintptr_t> dsts) {
+  std::vector<float*> d;
+  for (auto p : dsts) d.push_back(reinterpret_cast<float*>(p));
+  cpu_apply(static_cast<Codec>(codec), reinterpret_cast<uint8_t*>(payload), n,
+            static_cast<float>(scale), d.data(), static_cast<int>(d.size()));
+}
+
+double py_cpu_scale(int codec, uintptr_t delta, int64_t n, int stride) {
+  return cpu_compute_scale(static_cast<Codec>(codec),
+                           reinterpret_cast<float*>(delta), n, stride);
+}
+
+}  // namespace
+
+PYBIND11_MODULE(_core, m) {
+  m.doc() = "MI355X-native shared-tensor engine (CDNA4 HIP + RCCL/TCP)";
+
+  py::class_<Config>(m, "Config")
+      .def(py::init<>())
+      .def_readwrite("host", &Config::host)
+      .def_readwrite("port", &Config::port)
+      .def_readwrite("device", &Config::device)
+      .def_property(
+          "codec", [](const Config& c) { return static_cast<int>(c.codec); },
+          [](Config& c, int v) { c.codec = static_cast<Codec>(v); })
+      .def_readwrite("snapshot_join", &Config::snapshot_join)
+      .def_readwrite("use_rccl", &Config::use_rccl)
+      .def_readwrite("reconnect", &Config::reconnect)
+      .def_readwrite("keepalive_s", &Config::keepalive_s)
+      .def_readwrite("bw_limit", &Config::bw_limit)
+      .def_readwrite("expected_children", &Config::expected_children)
+      .def_readwrite("sizes", &Config::sizes)
+      .def_readwrite("explicit_parent", &Config::explicit_parent)
+      .def_readwrite("join_timeout_s", &Config::join_timeout_s)
+      .def_readwrite("rms_sample_stride", &Config::rms_sample_stride);
+
+  py::class_<Engine>(m, "Engine")
+      .def(py::init<Config>())
+      .def("set_values", &Engine::set_values)
+      .def("set_link_buffers", &Engine::set_link_buffers)
+      .def("start", &Engine::start, py::call_guard<py::gil_scoped_release>())
+      .def("add_from", &Engine::add_from,
+           py::call_guard<py::gil_scoped_release>())
+      .def("copy_to", &Engine::copy_to,
+           py::call_guard<py::gil_scoped_release>())
+      .def("fused_sgd", &Engine::fused_sgd,
+           py::call_guard<py::gil_scoped_release>())
+      .def("notify_dirty", &Engine::notify_dirty)
+      .def("close", &Engine::close, py::call_guard<py::gil_scoped_release>())
+      .def("is_master", &Engine::is_master)
+      .def("listen_port", &Engine::listen_port)
+      .def("last_error", &Engine::last_error)
+      .def("recent_scales_sent", &Engine::recent_scales_sent)
+      .def("recent_scales_recv", &Engine::recent_scales_recv)
+      .def("link_stats", [](Engine& e) {
+        py::list out;
+        for (auto& s : e.link_stats()) {
+          py::dict d;
+          d["rounds_sent"] = s.rounds_sent;
+          d["rounds_recv"] = s.rounds_recv;
+          d["bytes_sent"] = s.bytes_sent;
+          d["bytes_recv"] = s.bytes_recv;
+          d["last_scale_sent"] = s.last_scale_sent;
+          d["last_scale_recv"] = s.last_scale_recv;
+          d["active"] = s.active;
+          d["dead"] = s.dead;
+          d["peer"] = s.peer;
+          d["rccl"] = s.rccl;
+          out.append(d);
+        }
+        return out;
+      });
+
+  m.def("msg_bytes", &Engine::msg_bytes);
+  m.def("scales_area", &Engine::scales_area);
+  m.def("payload_bytes",
+        [](int c, int64_t n) { return payload_bytes(static_cast<Codec>(c), n); });
+  m.def("pad64", &pad64);
+  m.def("cpu_encode", &py_cpu_encode);
+  m.def("cpu_apply", &py_cpu_apply);
+  m.def("cpu_scale", &py_cpu_scale);
+  m.def("f32_to_e4m3", [](float x) { return static_cast<int>(f32_to_e4m3(x)); });
+  m.def("e4m3_to_f32", [](int v) { return e4m3_to_f32(static_cast<uint8_t>(v)); });
+
+#ifdef SHAMD_WITH_HIP
+  m.attr("with_hip") = true;
+#else
+  m.attr("with_hip") = false;
+#endif
+}

@@ -1,0 +1,165 @@
+#include "codec_cpu.h"
+
+#include <cmath>
+#include <cstring>
+
+namespace shamd {
+
+// ----------------------------------------------------------------- fp8 e4m3
+uint8_t f32_to_e4m3(float xf) {
+  if (xf != xf) return 0x7F;  // NaN
+  float cl = xf > 448.f ? 448.f : (xf < -448.f ? -448.f : xf);
+  uint32_t u;
+  std::memcpy(&u, &cl, 4);
+  uint8_t sign = (u >> 24) & 0x80;
+  uint32_t abs = u & 0x7FFFFFFFu;
+  if (abs == 0) return sign;
+  int e = static_cast<int>(abs >> 23) - 127;
+  uint32_t mant = (abs & 0x7FFFFFu) | 0x800000u;  // 24-bit with implicit bit
+  // normals keep 3 mantissa bits (drop 20); subnormal targets are multiples
+  // of 2^-9, needing a deeper shift for smaller exponents
+  int shift = (e >= -6) ? 20 : (14 - e);
+  if (shift > 24) return sign;  // |x| < 2^-10 half-quantum: rounds to zero
+  uint32_t q = mant >> shift;
+  uint32_t rem = mant & ((1u << shift) - 1);
+  uint32_t half = 1u << (shift - 1);
+  if (rem > half || (rem == half && (q & 1))) q++;
+  if (e >= -6) {
+    if (q == 16) { q = 8; e++; }  // mantissa carry
+    return sign | static_cast<uint8_t>(((e + 7) << 3) | (q - 8));
+  }
+  if (q >= 8) return sign | 0x08;  // rounded up to the smallest normal 2^-6
+  return sign | static_cast<uint8_t>(q);
+}
+
+float e4m3_to_f32(uint8_t v) {
+  int sign = (v & 0x80) ? -1 : 1;
+  uint32_t exp = (v >> 3) & 0xF;
+  uint32_t man = v & 0x7;
+  if (exp == 0xF && man == 0x7) return NAN;  // e4m3fn NaN
+  if (exp == 0) return sign * std::ldexp(static_cast<float>(man), -9);
+  return sign * std::ldexp(8.0f + man, static_cast<int>(exp) - 10);
+}
+
+// ------------------------------------------------------------ pow2 helpers
+float pow2_floor_f(double x) {
+  if (!(x > 0.0) || std::isinf(x) || std::isnan(x)) return 0.0f;
+  int e;
+  std::frexp(x, &e);  // x = m * 2^e, m in [0.5, 1)
+  return static_cast<float>(std::ldexp(1.0, e - 1));
+}
+
+float pow2_ceil_f(double x) {
+  if (!(x > 0.0) || std::isinf(x) || std::isnan(x)) return 0.0f;
+  int e;
+  double m = std::frexp(x, &e);
+  return static_cast<float>(std::ldexp(1.0, m == 0.5 ? e - 1 : e));
+}
+
+// ------------------------------------------------------------------ scales
+float cpu_compute_scale(Codec c, const float* delta, int64_t n, int stride) {
+  if (n <= 0) return 0.0f;
+  if (stride < 1) stride = 1;
+  if (c == Codec::OneBit) {
+    double ss = 0.0;
+    int64_t cnt = 0;
+    for (int64_t i = 0; i < n; i += stride, ++cnt) {
+      double d = atomic_load_f32(delta + i);
+      ss += d * d;
+    }
+    return pow2_floor_f(std::sqrt(ss / static_cast<double>(cnt)));
+  }
+  float mx = 0.0f;
+  for (int64_t i = 0; i < n; i += stride) {
+    float a = std::fabs(atomic_load_f32(delta + i));
+    if (a > mx) mx = a;
+  }
+  if (mx == 0.0f || std::isnan(mx) || std::isinf(mx)) return 0.0f;
+  return pow2_ceil_f(mx / (c == Codec::Fp8 ? 448.0 : 7.0));
+}
+
+// ---------------------------------------------------------------- quantize
+void cpu_quantize(Codec c, float* delta, int64_t n, float scale, uint8_t* payload) {
+  int64_t pb = payload_bytes(c, n);
+  std::memset(payload, 0, pb);
+  if (scale == 0.0f) return;
+  switch (c) {
+    case Codec::OneBit:
+      // residual > 0 -> bit 0, send +scale; else bit 1, send -scale
+      // (sharedtensor.c:166-174); debit with an atomic so adds that land
+      // between the read and the update are never lost.
+      for (int64_t i = 0; i < n; ++i) {
+        float v = atomic_load_f32(delta + i);
+        float sent;
+        if (v > 0.0f) {
+          sent = scale;
+        } else {
+          payload[i / 8] |= 1u << (i % 8);
+          sent = -scale;
+        }
+        atomic_add_f32(delta + i, -sent);
+      }
+      break;
+    case Codec::Fp8: {
+      float inv = 1.0f / scale;
+      for (int64_t i = 0; i < n; ++i) {
+        float v = atomic_load_f32(delta + i);
+        float sc = v * inv;
+        sc = sc > 448.f ? 448.f : (sc < -448.f ? -448.f : sc);
+        uint8_t q = f32_to_e4m3(sc);
+        payload[i] = q;
+        atomic_add_f32(delta + i, -e4m3_to_f32(q) * scale);
+      }
+      break;
+    }
+    case Codec::Int4: {
+      float inv = 1.0f / scale;
+      for (int64_t i = 0; i < n; ++i) {
+        float v = atomic_load_f32(delta + i);
+        float r = std::nearbyintf(v * inv);  // round-to-nearest-even
+        r = r > 7.f ? 7.f : (r < -7.f ? -7.f : r);
+        int8_t q = static_cast<int8_t>(r);
+        payload[i / 2] |= static_cast<uint8_t>(q & 0xF) << ((i % 2) * 4);
+        atomic_add_f32(delta + i, -static_cast<float>(q) * scale);
+      }
+      break;
+    }
+  }
+}
+
+// ------------------------------------------------------------------- apply
+void cpu_apply(Codec c, const uint8_t* payload, int64_t n, float scale,
+               float* const* dsts, int ndst) {
+  if (scale == 0.0f) return;
+  for (int64_t i = 0; i < n; ++i) {
+    float v = 0.0f;
+    switch (c) {
+      case Codec::OneBit: {
+        int bit = (payload[i / 8] >> (i % 8)) & 1;
+        v = bit ? -scale : scale;  // save_deltas, sharedtensor.c:106-111
+        break;
+      }
+      case Codec::Fp8:
+        v = e4m3_to_f32(payload[i]) * scale;
+        break;
+      case Codec::Int4: {
+        int8_t q = static_cast<int8_t>((payload[i / 2] >> ((i % 2) * 4)) & 0xF);
+        if (q > 7) q -= 16;
+        v = static_cast<float>(q) * scale;
+        break;
+      }
+    }
+    if (v != 0.0f)
+      for (int d = 0; d < ndst; ++d) atomic_add_f32(dsts[d] + i, v);
+  }
+}
+
+void cpu_add_scatter(const float* src, int64_t n, float* const* dsts, int ndst) {
+  for (int64_t i = 0; i < n; ++i) {
+    float v = src[i];
+    if (v != 0.0f)
+      for (int d = 0; d < ndst; ++d) atomic_add_f32(dsts[d] + i, v);
+  }
+}
+
+}  // namespace shamd

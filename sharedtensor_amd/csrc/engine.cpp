@@ -1,0 +1,1029 @@
+#include "engine.h"
+
+#include <arpa/inet.h>
+#include <fcntl.h>
+#include <netdb.h>
+#include <netinet/tcp.h>
+#include <sys/socket.h>
+#include <unistd.h>
+
+#include <chrono>
+#include <cmath>
+#include <cstdio>
+#include <cstring>
+#include <functional>
+#include <stdexcept>
+
+#include "codec_cpu.h"
+#include "rccl_transport.h"
+
+namespace shamd {
+
+using Clock = std::chrono::steady_clock;
+
+#define HIP_TRY(expr)                                                        \
+  do {                                                                       \
+    hipError_t _e = (expr);                                                  \
+    if (_e != hipSuccess)                                                    \
+      throw std::runtime_error(std::string("HIP error: ") +                  \
+                               hipGetErrorString(_e) + " @" __FILE__ ":" +   \
+                               std::to_string(__LINE__));                    \
+  } while (0)
+
+// ------------------------------------------------------------ I/O helpers
+// Loop partial reads/writes and retry EINTR (reference read_or_die/
+// write_or_die, sharedtensor.c:53-87) — but surface failure instead of
+// exiting the process.
+static bool io_read(int fd, void* buf, size_t count) {
+  auto* p = static_cast<uint8_t*>(buf);
+  while (count) {
+    ssize_t r = ::read(fd, p, count);
+    if (r == 0) return false;  // peer closed
+    if (r < 0) {
+      if (errno == EINTR) continue;
+      return false;
+    }
+    count -= static_cast<size_t>(r);
+    p += r;
+  }
+  return true;
+}
+
+static bool io_write(int fd, const void* buf, size_t count) {
+  auto* p = static_cast<const uint8_t*>(buf);
+  while (count) {
+    ssize_t r = ::write(fd, p, count);
+    if (r < 0) {
+      if (errno == EINTR) continue;
+      return false;
+    }
+    count -= static_cast<size_t>(r);
+    p += r;
+  }
+  return true;
+}
+
+static void set_sockopts(int fd) {
+  int yes = 1;
+  setsockopt(fd, IPPROTO_TCP, TCP_NODELAY, &yes, sizeof(yes));
+  // on the up-connection this is what later allows binding our listener to
+  // the same local (ip, port) — the reference's self-addressing trick
+  // (sharedtensor.c:264,292-316)
+  setsockopt(fd, SOL_SOCKET, SO_REUSEADDR, &yes, sizeof(yes));
+  int buf = 8 << 20;
+  setsockopt(fd, SOL_SOCKET, SO_SNDBUF, &buf, sizeof(buf));
+  setsockopt(fd, SOL_SOCKET, SO_RCVBUF, &buf, sizeof(buf));
+}
+
+static bool resolve_ipv4(const std::string& host, int port, sockaddr_in* out) {
+  std::memset(out, 0, sizeof(*out));
+  out->sin_family = AF_INET;
+  out->sin_port = htons(static_cast<uint16_t>(port));
+  addrinfo hints{};
+  hints.ai_family = AF_INET;
+  hints.ai_socktype = SOCK_STREAM;
+  addrinfo* res = nullptr;
+  if (getaddrinfo(host.c_str(), nullptr, &hints, &res) != 0 || !res) return false;
+  out->sin_addr = reinterpret_cast<sockaddr_in*>(res->ai_addr)->sin_addr;
+  freeaddrinfo(res);
+  return true;
+}
+
+static std::string addr_str(const sockaddr_in& a) {
+  char b[64];
+  inet_ntop(AF_INET, &a.sin_addr, b, sizeof(b));
+  return std::string(b) + ":" + std::to_string(ntohs(a.sin_port));
+}
+
+static uint64_t compute_hostid() {
+  // same-host detection for the RCCL/xGMI upgrade
+  uint64_t h = 1469598103934665603ull;
+  auto mix = [&h](const char* s, size_t n) {
+    for (size_t i = 0; i < n; ++i) {
+      h ^= static_cast<uint8_t>(s[i]);
+      h *= 1099511628211ull;
+    }
+  };
+  FILE* f = std::fopen("/etc/machine-id", "r");
+  char buf[128] = {0};
+  if (f) {
+    size_t n = std::fread(buf, 1, sizeof(buf), f);
+    std::fclose(f);
+    mix(buf, n);
+  }
+  char hn[256] = {0};
+  gethostname(hn, sizeof(hn) - 1);
+  mix(hn, std::strlen(hn));
+  return h;
+}
+
+// ----------------------------------------------------------------- Engine
+
+int64_t Engine::scales_area(const Config& cfg) {
+  return align8(4 * static_cast<int64_t>(cfg.sizes.size()));
+}
+
+int64_t Engine::msg_bytes(const Config& cfg) {
+  int64_t p = 0;
+  for (int64_t s : cfg.sizes) p += payload_bytes(cfg.codec, s);
+  return scales_area(cfg) + p;
+}
+
+Engine::Engine(Config cfg) : cfg_(std::move(cfg)) {
+  if (cfg_.sizes.empty()) throw std::runtime_error("sizes must be non-empty");
+  T_ = static_cast<int>(cfg_.sizes.size());
+  offs_.resize(T_ + 1);
+  poffs_.resize(T_ + 1);
+  offs_[0] = poffs_[0] = 0;
+  for (int t = 0; t < T_; ++t) {
+    offs_[t + 1] = offs_[t] + cfg_.sizes[t];
+    poffs_[t + 1] = poffs_[t] + pad64(cfg_.sizes[t]);
+  }
+  n_ = offs_[T_];
+  pe_ = poffs_[T_];
+  SA_ = scales_area(cfg_);
+  P_ = msg_bytes(cfg_) - SA_;
+  hostid_ = compute_hostid();
+  for (int i = 0; i < 3; ++i) links_[i].idx = i;
+}
+
+Engine::~Engine() {
+  try {
+    close();
+  } catch (...) {
+  }
+}
+
+void Engine::set_values(uintptr_t p) { values_ = reinterpret_cast<float*>(p); }
+
+void Engine::set_link_buffers(int link, uintptr_t delta, uintptr_t send_buf,
+                              uintptr_t recv_buf, uintptr_t send_pin,
+                              uintptr_t recv_pin) {
+  Link& lk = links_[link];
+  lk.delta = reinterpret_cast<float*>(delta);
+  lk.send_buf = reinterpret_cast<uint8_t*>(send_buf);
+  lk.recv_buf = reinterpret_cast<uint8_t*>(recv_buf);
+  lk.send_pin = reinterpret_cast<uint8_t*>(send_pin);
+  lk.recv_pin = reinterpret_cast<uint8_t*>(recv_pin);
+  lk.provisioned = lk.delta != nullptr;
+}
+
+void Engine::set_error(const std::string& e) {
+  std::lock_guard<std::mutex> g(err_m_);
+  last_error_ = e;
+}
+
+std::string Engine::last_error() {
+  std::lock_guard<std::mutex> g(err_m_);
+  return last_error_;
+}
+
+void Engine::push_scale(bool sent, float s) {
+  std::lock_guard<std::mutex> g(ring_m_);
+  auto& r = sent ? ring_sent_ : ring_recv_;
+  r.push_back(s);
+  if (r.size() > 2048) r.pop_front();
+}
+
+std::vector<float> Engine::recent_scales_sent() {
+  std::lock_guard<std::mutex> g(ring_m_);
+  return {ring_sent_.begin(), ring_sent_.end()};
+}
+
+std::vector<float> Engine::recent_scales_recv() {
+  std::lock_guard<std::mutex> g(ring_m_);
+  return {ring_recv_.begin(), ring_recv_.end()};
+}
+
+// ------------------------------------------------------------------- GPU
+
+void Engine::init_gpu() {
+  if (!gpu()) return;
+  HIP_TRY(hipSetDevice(cfg_.device));
+  HIP_TRY(hipMalloc(&dtb_.offs, sizeof(int64_t) * (T_ + 1) * 2));
+  dtb_.poffs = dtb_.offs + (T_ + 1);
+  HIP_TRY(hipMemcpy(dtb_.offs, offs_.data(), sizeof(int64_t) * (T_ + 1),
+                    hipMemcpyHostToDevice));
+  HIP_TRY(hipMemcpy(dtb_.poffs, poffs_.data(), sizeof(int64_t) * (T_ + 1),
+                    hipMemcpyHostToDevice));
+  dtb_.T = T_;
+  dtb_.n = n_;
+  dtb_.pe = pe_;
+  for (auto& lk : links_) {
+    if (!lk.provisioned) continue;
+    HIP_TRY(hipMalloc(&lk.reduce_buf, 8 * T_));
+    HIP_TRY(hipStreamCreateWithFlags(&lk.s_send, hipStreamNonBlocking));
+    HIP_TRY(hipStreamCreateWithFlags(&lk.s_recv, hipStreamNonBlocking));
+  }
+}
+
+void Engine::free_gpu() {
+  if (!gpu()) return;
+  for (auto& lk : links_) {
+    if (lk.s_send) hipStreamDestroy(lk.s_send), lk.s_send = nullptr;
+    if (lk.s_recv) hipStreamDestroy(lk.s_recv), lk.s_recv = nullptr;
+    if (lk.reduce_buf) hipFree(lk.reduce_buf), lk.reduce_buf = nullptr;
+  }
+  if (dtb_.offs) hipFree(dtb_.offs), dtb_.offs = nullptr;
+}
+
+// ------------------------------------------------------------------ start
+
+void Engine::start() {
+  if (started_) throw std::runtime_error("engine already started");
+  if (!values_) throw std::runtime_error("values buffer not set");
+  init_gpu();
+  join_tree();
+  started_ = true;
+  listen_thread_ = std::thread([this] { listen_loop(); });
+}
+
+bool Engine::try_connect(const sockaddr_in& addr, int& out_fd) {
+  int fd = ::socket(AF_INET, SOCK_STREAM, 0);
+  if (fd < 0) return false;
+  set_sockopts(fd);
+  if (::connect(fd, reinterpret_cast<const sockaddr*>(&addr), sizeof(addr)) < 0) {
+    ::close(fd);
+    return false;
+  }
+  out_fd = fd;
+  return true;
+}
+
+void Engine::join_tree() {
+  if (!resolve_ipv4(cfg_.host, cfg_.port, &root_addr_))
+    throw std::runtime_error("cannot resolve host " + cfg_.host);
+
+  auto deadline = Clock::now() + std::chrono::duration_cast<Clock::duration>(
+                                     std::chrono::duration<double>(cfg_.join_timeout_s));
+
+  sockaddr_in target = root_addr_;
+  bool explicit_mode = !cfg_.explicit_parent.empty();
+  if (explicit_mode) {
+    auto pos = cfg_.explicit_parent.rfind(':');
+    if (pos == std::string::npos)
+      throw std::runtime_error("explicit_parent must be ip:port");
+    if (!resolve_ipv4(cfg_.explicit_parent.substr(0, pos),
+                      std::stoi(cfg_.explicit_parent.substr(pos + 1)), &target))
+      throw std::runtime_error("cannot resolve explicit_parent");
+  }
+
+  int hops_since_root = 0;
+  while (true) {
+    if (Clock::now() > deadline) throw std::runtime_error("join timed out");
+    int fd = -1;
+    if (!try_connect(target, fd)) {
+      if (!explicit_mode && hops_since_root == 0) {
+        // nobody is listening at the rendezvous: we are the master
+        // (sharedtensor.c:271-277,318-322).  If another process won the race
+        // to bind, retry the walk instead of dying.
+        try {
+          become_master();
+          return;
+        } catch (const std::exception&) {
+          std::this_thread::sleep_for(std::chrono::milliseconds(100));
+          continue;
+        }
+      }
+      // a mid-walk node (or the explicit parent) is not up yet: back off
+      std::this_thread::sleep_for(std::chrono::milliseconds(200));
+      if (!explicit_mode) {
+        target = root_addr_;
+        hops_since_root = 0;
+      }
+      continue;
+    }
+    Hello h{};
+    h.magic = MAGIC;
+    h.version = PROTO_VERSION;
+    h.flags = static_cast<uint16_t>((gpu() ? HELLO_HAS_GPU : 0) |
+                                    (gpu() && cfg_.use_rccl ? HELLO_WANT_RCCL : 0));
+    h.n = static_cast<uint64_t>(n_);
+    h.ntensors = static_cast<uint32_t>(T_);
+    h.codec = static_cast<uint32_t>(cfg_.codec);
+    h.hostid = hostid_;
+    h.device = cfg_.device;
+    uint8_t reply = 0;
+    if (!io_write(fd, &h, sizeof(h)) || !io_read(fd, &reply, 1)) {
+      ::close(fd);
+      std::this_thread::sleep_for(std::chrono::milliseconds(100));
+      if (!explicit_mode) { target = root_addr_; hops_since_root = 0; }
+      continue;
+    }
+    if (reply == 'N') {
+      // walk down the tree (sharedtensor.c:298-300)
+      uint8_t buf[6];
+      if (!io_read(fd, buf, 6)) {
+        ::close(fd);
+        target = root_addr_;
+        hops_since_root = 0;
+        continue;
+      }
+      ::close(fd);
+      std::memcpy(&target.sin_addr.s_addr, buf, 4);
+      std::memcpy(&target.sin_port, buf + 4, 2);
+      hops_since_root++;
+      continue;
+    }
+    if (reply != 'Y') {
+      ::close(fd);
+      throw std::runtime_error("protocol error during join (reply byte " +
+                               std::to_string(int(reply)) + ")");
+    }
+    handshake_as_child(fd);
+    return;
+  }
+}
+
+void Engine::become_master() {
+  bind_listen(root_addr_);
+  is_master_ = true;
+  std::fprintf(stderr,
+               "[sharedtensor_amd] master tensor at %s (n=%lld, %d tensor%s)\n",
+               addr_str(root_addr_).c_str(), static_cast<long long>(n_), T_,
+               T_ == 1 ? "" : "s");
+}
+
+void Engine::handshake_as_child(int fd) {
+  AcceptHello ah{};
+  if (!io_read(fd, &ah, sizeof(ah)))
+    throw std::runtime_error("parent hung up during handshake");
+  if (ah.version != PROTO_VERSION || ah.n != static_cast<uint64_t>(n_) ||
+      ah.ntensors != static_cast<uint32_t>(T_))
+    throw std::runtime_error("handshake mismatch: parent has different tensor shape/version");
+  if (ah.codec != static_cast<uint32_t>(cfg_.codec))
+    throw std::runtime_error("handshake mismatch: parent uses a different codec");
+
+  Link& up = links_[LK_UP];
+  if (!up.provisioned)
+    throw std::runtime_error("up link not provisioned but joining as child");
+  up.fd = fd;
+  socklen_t alen = sizeof(up.peer);
+  getpeername(fd, reinterpret_cast<sockaddr*>(&up.peer), &alen);
+  up.peer_desc = addr_str(up.peer);
+
+  // bind our listener to the local address of the up socket so our parent's
+  // view of us (getpeername) is also our listen address — the reference's
+  // self-addressing trick (sharedtensor.c:292-316)
+  sockaddr_in self{};
+  socklen_t slen = sizeof(self);
+  getsockname(fd, reinterpret_cast<sockaddr*>(&self), &slen);
+  bind_listen(self);
+
+  if (ah.flags & ACC_RCCL) {
+    uint8_t ids[2 * RCCL_ID_BYTES];
+    if (!io_read(fd, ids, sizeof(ids)))
+      throw std::runtime_error("failed to read RCCL ids from parent");
+    rccl_upgrade(up, ids, /*is_parent=*/false);
+  }
+  if (ah.flags & ACC_SNAPSHOT) recv_snapshot(fd);
+
+  up.state.store(L_ACTIVE);
+  spawn_link_threads(up);
+}
+
+void Engine::bind_listen(const sockaddr_in& addr) {
+  listen_fd_ = ::socket(AF_INET, SOCK_STREAM, 0);
+  if (listen_fd_ < 0) throw std::runtime_error("socket() failed");
+  int yes = 1;
+  setsockopt(listen_fd_, SOL_SOCKET, SO_REUSEADDR, &yes, sizeof(yes));
+  if (::bind(listen_fd_, reinterpret_cast<const sockaddr*>(&addr), sizeof(addr)) < 0) {
+    std::string err = std::strerror(errno);
+    ::close(listen_fd_);
+    listen_fd_ = -1;
+    throw std::runtime_error("bind(" + addr_str(addr) + ") failed: " + err);
+  }
+  if (::listen(listen_fd_, 16) < 0) {
+    ::close(listen_fd_);
+    listen_fd_ = -1;
+    throw std::runtime_error("listen() failed");
+  }
+  sockaddr_in got{};
+  socklen_t glen = sizeof(got);
+  getsockname(listen_fd_, reinterpret_cast<sockaddr*>(&got), &glen);
+  listen_port_ = ntohs(got.sin_port);
+}
+
+// ----------------------------------------------------------------- listen
+
+void Engine::listen_loop() {
+  int redirect_rr = 0;  // alternate redirects (sharedtensor.c:230)
+  while (!closing_) {
+    sockaddr_in peer{};
+    socklen_t plen = sizeof(peer);
+    int fd = ::accept(listen_fd_, reinterpret_cast<sockaddr*>(&peer), &plen);
+    if (closing_) {
+      if (fd >= 0) ::close(fd);
+      break;
+    }
+    if (fd < 0) {
+      if (errno == EINTR) continue;
+      break;  // listen socket shut down
+    }
+    set_sockopts(fd);
+    timeval tv{5, 0};
+    setsockopt(fd, SOL_SOCKET, SO_RCVTIMEO, &tv, sizeof(tv));
+    Hello h{};
+    if (!io_read(fd, &h, sizeof(h)) || h.magic != MAGIC ||
+        h.version != PROTO_VERSION) {
+      ::close(fd);
+      continue;
+    }
+    if (h.n != static_cast<uint64_t>(n_) ||
+        h.ntensors != static_cast<uint32_t>(T_) ||
+        h.codec != static_cast<uint32_t>(cfg_.codec)) {
+      set_error("rejected joiner with mismatched tensor shape/codec");
+      ::close(fd);
+      continue;
+    }
+    timeval tv0{0, 0};
+    setsockopt(fd, SOL_SOCKET, SO_RCVTIMEO, &tv0, sizeof(tv0));
+
+    // pick a free child slot
+    int slot = -1;
+    for (int i : {LK_LEFT, LK_RIGHT}) {
+      Link& lk = links_[i];
+      if (!lk.provisioned) continue;
+      int st = lk.state.load();
+      if (st == L_FREE) { slot = i; break; }
+      if (st == L_DEAD) {
+        // reconnection support (reference TODO, README.md:33): reclaim the
+        // slot once its old threads have exited
+        if (lk.t_send.joinable()) lk.t_send.join();
+        if (lk.t_recv.joinable()) lk.t_recv.join();
+        if (lk.t_ctrl.joinable()) lk.t_ctrl.join();
+        if (lk.fd >= 0) ::close(lk.fd);
+        if (lk.rccl_link) {
+          rccl_destroy(static_cast<RcclLink*>(lk.rccl_link));
+          lk.rccl_link = nullptr;
+          lk.rccl = false;
+        }
+        lk.abort.store(false);
+        lk.fd = -1;
+        lk.error.clear();
+        lk.state.store(L_FREE);
+        slot = i;
+        break;
+      }
+    }
+    if (slot < 0) {
+      // no capacity: redirect down the tree (sharedtensor.c:224-234)
+      std::vector<sockaddr_in> cands;
+      for (int i : {LK_LEFT, LK_RIGHT})
+        if (links_[i].state.load() == L_ACTIVE) cands.push_back(links_[i].peer);
+      sockaddr_in tgt;
+      if (!cands.empty()) {
+        tgt = cands[(redirect_rr++) % cands.size()];
+      } else if (links_[LK_UP].state.load() == L_ACTIVE) {
+        tgt = links_[LK_UP].peer;  // leaf with no child slots: bounce upward
+      } else {
+        sockaddr_in self{};
+        socklen_t sl = sizeof(self);
+        getsockname(listen_fd_, reinterpret_cast<sockaddr*>(&self), &sl);
+        tgt = self;
+      }
+      uint8_t msg[7];
+      msg[0] = 'N';
+      std::memcpy(msg + 1, &tgt.sin_addr.s_addr, 4);
+      std::memcpy(msg + 5, &tgt.sin_port, 2);
+      io_write(fd, msg, 7);
+      ::close(fd);
+      continue;
+    }
+    accept_child(fd, h, peer, slot);
+  }
+}
+
+void Engine::accept_child(int fd, const Hello& h, const sockaddr_in& peer,
+                          int slot) {
+  Link& lk = links_[slot];
+  lk.fd = fd;
+  lk.peer = peer;
+  lk.peer_desc = addr_str(peer);
+  uint8_t yes = 'Y';
+  bool upgrade = rccl_wanted(h);
+  AcceptHello ah{};
+  ah.version = PROTO_VERSION;
+  ah.flags = static_cast<uint16_t>((cfg_.snapshot_join ? ACC_SNAPSHOT : 0) |
+                                   (upgrade ? ACC_RCCL : 0));
+  ah.codec = static_cast<uint32_t>(cfg_.codec);
+  ah.n = static_cast<uint64_t>(n_);
+  ah.ntensors = static_cast<uint32_t>(T_);
+  if (!io_write(fd, &yes, 1) || !io_write(fd, &ah, sizeof(ah))) {
+    ::close(fd);
+    lk.fd = -1;
+    return;
+  }
+  if (upgrade) {
+    try {
+      uint8_t ids[2 * RCCL_ID_BYTES];
+      rccl_make_ids(ids);
+      if (!io_write(fd, ids, sizeof(ids)))
+        throw std::runtime_error("failed to send RCCL ids");
+      rccl_upgrade(lk, ids, /*is_parent=*/true);
+    } catch (const std::exception& e) {
+      set_error(std::string("rccl upgrade failed: ") + e.what());
+      ::close(fd);
+      lk.fd = -1;
+      return;
+    }
+  }
+  if (cfg_.snapshot_join) {
+    try {
+      send_snapshot(lk);
+    } catch (const std::exception& e) {
+      set_error(std::string("snapshot send failed: ") + e.what());
+      ::close(fd);
+      lk.fd = -1;
+      return;
+    }
+  }
+  lk.state.store(L_ACTIVE);
+  spawn_link_threads(lk);
+}
+
+void Engine::spawn_link_threads(Link& lk) {
+  lk.t_send = std::thread([this, &lk] { send_loop(lk); });
+  lk.t_recv = std::thread([this, &lk] { recv_loop(lk); });
+  if (lk.rccl) lk.t_ctrl = std::thread([this, &lk] { ctrl_loop(lk); });
+}
+
+bool Engine::rccl_wanted(const Hello& h) const {
+  return gpu() && cfg_.use_rccl && (h.flags & HELLO_WANT_RCCL) &&
+         h.hostid == hostid_ && h.device >= 0 && h.device != cfg_.device;
+}
+
+void Engine::rccl_upgrade(Link& lk, const uint8_t* ids, bool is_parent) {
+  lk.rccl_link = rccl_link_create(cfg_.device, ids, is_parent, cfg_.join_timeout_s);
+  lk.rccl = true;
+}
+
+// TCP control reader for RCCL-upgraded links: keepalives + close + death
+// detection while the payload flows over xGMI.
+void Engine::ctrl_loop(Link& lk) {
+  while (!closing_ && lk.state.load() == L_ACTIVE) {
+    PacketHeader hdr{};
+    if (!io_read(lk.fd, &hdr, 8)) {
+      link_down(lk, "peer disconnected (ctrl)", true);
+      break;
+    }
+    if (hdr.type == PKT_CLOSE) {
+      link_down(lk, "peer closed", true);
+      break;
+    }
+    // PING or anything else: ignore
+  }
+}
+
+// --------------------------------------------------------------- snapshot
+// v2 fast join: instead of the reference's bootstrap-by-accumulated-delta
+// (sharedtensor.c:379-388, O(dynamic range) gossip rounds), the parent
+// streams its current values once and debits exactly the bytes sent from the
+// link's delta buffer — preserving the error-feedback invariant under
+// concurrent updates.
+
+void Engine::send_snapshot(Link& lk) {
+  const int64_t chunk_bytes = gpu() ? (SA_ + P_) : (1 << 20);
+  const int64_t chunk_elems = std::max<int64_t>(chunk_bytes / 4, 1);
+  std::vector<uint8_t> tmp;
+  if (!gpu()) tmp.resize(static_cast<size_t>(chunk_elems) * 4);
+  for (int64_t off = 0; off < n_; off += chunk_elems) {
+    int64_t ce = std::min(chunk_elems, n_ - off);
+    if (gpu()) {
+      HIP_TRY(hipSetDevice(cfg_.device));
+      HIP_TRY(hipMemcpyAsync(lk.send_pin, values_ + off, ce * 4,
+                             hipMemcpyDeviceToHost, lk.s_send));
+      HIP_TRY(hipStreamSynchronize(lk.s_send));
+      if (!io_write(lk.fd, lk.send_pin, ce * 4))
+        throw std::runtime_error("tcp write failed");
+      // debit: delta -= exactly-what-was-sent
+      HIP_TRY(hipMemcpyAsync(lk.send_buf, lk.send_pin, ce * 4,
+                             hipMemcpyHostToDevice, lk.s_send));
+      hip_add_scatter(reinterpret_cast<float*>(lk.send_buf), ce, -1.0f,
+                      lk.delta + off, nullptr, nullptr, nullptr, lk.s_send);
+      HIP_TRY(hipStreamSynchronize(lk.s_send));
+    } else {
+      for (int64_t i = 0; i < ce; ++i)
+        reinterpret_cast<float*>(tmp.data())[i] = atomic_load_f32(values_ + off + i);
+      if (!io_write(lk.fd, tmp.data(), ce * 4))
+        throw std::runtime_error("tcp write failed");
+      const float* snap = reinterpret_cast<const float*>(tmp.data());
+      for (int64_t i = 0; i < ce; ++i)
+        if (snap[i] != 0.0f) atomic_add_f32(lk.delta + off + i, -snap[i]);
+    }
+    lk.bytes_sent += ce * 4;
+  }
+}
+
+void Engine::recv_snapshot(int fd) {
+  Link& up = links_[LK_UP];
+  const int64_t chunk_bytes = gpu() ? (SA_ + P_) : (1 << 20);
+  const int64_t chunk_elems = std::max<int64_t>(chunk_bytes / 4, 1);
+  std::vector<uint8_t> tmp;
+  if (!gpu()) tmp.resize(static_cast<size_t>(chunk_elems) * 4);
+  float* fwd[2] = {
+      links_[LK_LEFT].provisioned ? links_[LK_LEFT].delta : nullptr,
+      links_[LK_RIGHT].provisioned ? links_[LK_RIGHT].delta : nullptr};
+  for (int64_t off = 0; off < n_; off += chunk_elems) {
+    int64_t ce = std::min(chunk_elems, n_ - off);
+    if (gpu()) {
+      HIP_TRY(hipSetDevice(cfg_.device));
+      if (!io_read(fd, up.recv_pin, ce * 4))
+        throw std::runtime_error("snapshot read failed");
+      HIP_TRY(hipMemcpyAsync(up.recv_buf, up.recv_pin, ce * 4,
+                             hipMemcpyHostToDevice, up.s_recv));
+      // values += snap; future children's deltas += snap (join-state
+      // forwarding, the GPU analog of sharedtensor.c:379-381)
+      hip_add_scatter(reinterpret_cast<float*>(up.recv_buf), ce, 1.0f,
+                      values_ + off, fwd[0] ? fwd[0] + off : nullptr,
+                      fwd[1] ? fwd[1] + off : nullptr, nullptr, up.s_recv);
+      HIP_TRY(hipStreamSynchronize(up.s_recv));
+    } else {
+      if (!io_read(fd, tmp.data(), ce * 4))
+        throw std::runtime_error("snapshot read failed");
+      const float* snap = reinterpret_cast<const float*>(tmp.data());
+      for (int64_t i = 0; i < ce; ++i) {
+        float v = snap[i];
+        if (v == 0.0f) continue;
+        atomic_add_f32(values_ + off + i, v);
+        if (fwd[0]) atomic_add_f32(fwd[0] + off + i, v);
+        if (fwd[1]) atomic_add_f32(fwd[1] + off + i, v);
+      }
+    }
+    up.bytes_recv += ce * 4;
+  }
+}
+
+// -------------------------------------------------------------- send side
+
+void Engine::compute_scales(Link& lk, float* scales_host) {
+  if (gpu()) {
+    HIP_TRY(hipSetDevice(cfg_.device));
+    hip_reduce_scales(cfg_.codec, lk.delta, dtb_, lk.reduce_buf,
+                      reinterpret_cast<float*>(lk.send_buf),
+                      cfg_.rms_sample_stride, lk.s_send);
+    HIP_TRY(hipMemcpyAsync(scales_host, lk.send_buf, 4 * T_,
+                           hipMemcpyDeviceToHost, lk.s_send));
+    HIP_TRY(hipStreamSynchronize(lk.s_send));
+  } else {
+    for (int t = 0; t < T_; ++t)
+      scales_host[t] = cpu_compute_scale(cfg_.codec, lk.delta + offs_[t],
+                                         cfg_.sizes[t], cfg_.rms_sample_stride);
+    std::memcpy(lk.send_pin + 8, scales_host, 4 * T_);
+  }
+}
+
+bool Engine::send_packet(Link& lk, const float* scales_host) {
+  PacketHeader hdr{};
+  hdr.type = PKT_DATA;
+  hdr.codec = static_cast<uint8_t>(cfg_.codec);
+  hdr.ntensors = static_cast<uint32_t>(T_);
+  if (gpu()) {
+    std::memcpy(lk.send_pin, &hdr, 8);
+    HIP_TRY(hipMemcpyAsync(lk.send_pin + 8, lk.send_buf, SA_ + P_,
+                           hipMemcpyDeviceToHost, lk.s_send));
+    HIP_TRY(hipStreamSynchronize(lk.s_send));
+  } else {
+    std::memcpy(lk.send_pin, &hdr, 8);
+  }
+  std::lock_guard<std::mutex> g(lk.wm);
+  if (!io_write(lk.fd, lk.send_pin, 8 + SA_ + P_)) return false;
+  return true;
+}
+
+void Engine::send_loop(Link& lk) {
+  if (gpu()) hipSetDevice(cfg_.device);
+  std::vector<float> scales(T_);
+  auto last_send = Clock::now();
+  auto next_allowed = Clock::now();
+  const double keepalive = cfg_.keepalive_s;
+  while (!closing_ && lk.state.load() == L_ACTIVE) {
+    try {
+      compute_scales(lk, scales.data());
+    } catch (const std::exception& e) {
+      link_down(lk, std::string("scale reduction failed: ") + e.what(), false);
+      break;
+    }
+    float maxs = 0.f;
+    for (float s : scales) maxs = std::max(maxs, s);
+    if (maxs == 0.0f) {
+      // idle: cheap PING instead of the reference's full zero packet
+      // (sharedtensor.c:161-177); wake instantly when a delta lands
+      auto now = Clock::now();
+      if (std::chrono::duration<double>(now - last_send).count() >= keepalive) {
+        PacketHeader ping{};
+        ping.type = PKT_PING;
+        ping.ntensors = static_cast<uint32_t>(T_);
+        std::lock_guard<std::mutex> g(lk.wm);
+        if (!io_write(lk.fd, &ping, 8)) {
+          link_down(lk, "keepalive write failed", false);
+          break;
+        }
+        last_send = now;
+      }
+      std::unique_lock<std::mutex> l(lk.m);
+      lk.cv.wait_for(l, std::chrono::duration<double>(keepalive), [&] {
+        return lk.dirty || closing_ || lk.state.load() != L_ACTIVE;
+      });
+      lk.dirty = false;
+      continue;
+    }
+    // bandwidth cap (reference TODO, README.md:31): token-bucket pacing
+    if (cfg_.bw_limit > 0) {
+      auto now = Clock::now();
+      if (now < next_allowed)
+        std::this_thread::sleep_for(next_allowed - now);
+      next_allowed = Clock::now() + std::chrono::duration_cast<Clock::duration>(
+          std::chrono::duration<double>((8.0 + SA_ + P_) / cfg_.bw_limit));
+    }
+    try {
+      if (gpu()) {
+        hip_quantize(cfg_.codec, lk.delta, dtb_,
+                     reinterpret_cast<float*>(lk.send_buf), lk.send_buf + SA_,
+                     lk.s_send);
+      } else {
+        for (int t = 0; t < T_; ++t)
+          cpu_quantize(cfg_.codec, lk.delta + offs_[t], cfg_.sizes[t], scales[t],
+                       lk.send_pin + 8 + SA_ +
+                           (poffs_[t] / 64) * (payload_bytes(cfg_.codec, 64)));
+      }
+    } catch (const std::exception& e) {
+      link_down(lk, std::string("quantize failed: ") + e.what(), false);
+      break;
+    }
+    if (lk.rccl) {
+      // xGMI data plane: the packed message goes device-to-device
+      if (!rccl_send(static_cast<RcclLink*>(lk.rccl_link), lk.send_buf,
+                     SA_ + P_, lk.s_send, lk.abort)) {
+        link_down(lk, "rccl send failed", lk.abort.load());
+        break;
+      }
+    } else if (!send_packet(lk, scales.data())) {
+      link_down(lk, "data write failed", false);
+      break;
+    }
+    last_send = Clock::now();
+    lk.rounds_sent++;
+    lk.bytes_sent += 8 + SA_ + P_;
+    lk.last_scale_sent.store(maxs);
+    push_scale(true, maxs);
+  }
+}
+
+// -------------------------------------------------------------- recv side
+
+void Engine::apply_packet(Link& lk, const float* scales_host) {
+  // destinations: local replica + gossip-forward into the other links'
+  // delta buffers, excluding the source (sharedtensor.c:124-127)
+  float* fwd[2] = {nullptr, nullptr};
+  int nf = 0;
+  for (int i = 0; i < 3; ++i)
+    if (i != lk.idx && links_[i].provisioned) fwd[nf++] = links_[i].delta;
+  if (gpu()) {
+    HIP_TRY(hipSetDevice(cfg_.device));
+    if (!lk.rccl)  // TCP staging; RCCL already delivered into recv_buf
+      HIP_TRY(hipMemcpyAsync(lk.recv_buf, lk.recv_pin + 8, SA_ + P_,
+                             hipMemcpyHostToDevice, lk.s_recv));
+    hip_apply(cfg_.codec, lk.recv_buf + SA_, dtb_,
+              reinterpret_cast<float*>(lk.recv_buf), values_, fwd[0], fwd[1],
+              nullptr, lk.s_recv);
+    HIP_TRY(hipStreamSynchronize(lk.s_recv));
+  } else {
+    for (int t = 0; t < T_; ++t) {
+      float* dsts[3] = {values_ + offs_[t],
+                        fwd[0] ? fwd[0] + offs_[t] : nullptr,
+                        fwd[1] ? fwd[1] + offs_[t] : nullptr};
+      int nd = 1 + (fwd[0] ? 1 : 0) + (fwd[1] ? 1 : 0);
+      cpu_apply(cfg_.codec,
+                lk.recv_pin + 8 + SA_ +
+                    (poffs_[t] / 64) * (payload_bytes(cfg_.codec, 64)),
+                cfg_.sizes[t], scales_host[t], dsts, nd);
+    }
+  }
+}
+
+void Engine::recv_loop(Link& lk) {
+  if (gpu()) hipSetDevice(cfg_.device);
+  std::vector<float> scales(T_);
+  while (!closing_ && lk.state.load() == L_ACTIVE) {
+    if (lk.rccl) {
+      if (!rccl_recv(static_cast<RcclLink*>(lk.rccl_link), lk.recv_buf,
+                     SA_ + P_, lk.s_recv, lk.abort)) {
+        link_down(lk, "rccl recv ended", true);
+        break;
+      }
+      try {
+        HIP_TRY(hipMemcpy(scales.data(), lk.recv_buf, 4 * T_,
+                          hipMemcpyDeviceToHost));
+        float maxs = 0.f;
+        for (float s : scales) maxs = std::max(maxs, s);
+        lk.rounds_recv++;
+        lk.bytes_recv += SA_ + P_;
+        lk.last_scale_recv.store(maxs);
+        push_scale(false, maxs);
+        if (maxs != 0.0f) apply_packet(lk, scales.data());
+      } catch (const std::exception& e) {
+        link_down(lk, std::string("apply failed: ") + e.what(), false);
+        break;
+      }
+      for (int i = 0; i < 3; ++i) {
+        if (i == lk.idx) continue;
+        Link& o = links_[i];
+        if (o.state.load() == L_ACTIVE) {
+          std::lock_guard<std::mutex> g(o.m);
+          o.dirty = true;
+          o.cv.notify_all();
+        }
+      }
+      continue;
+    }
+    PacketHeader hdr{};
+    if (!io_read(lk.fd, &hdr, 8)) {
+      link_down(lk, "peer disconnected", true);
+      break;
+    }
+    if (hdr.type == PKT_CLOSE) {
+      link_down(lk, "peer closed", true);
+      break;
+    }
+    if (hdr.type == PKT_PING) continue;
+    if (hdr.type != PKT_DATA ||
+        hdr.ntensors != static_cast<uint32_t>(T_) ||
+        hdr.codec != static_cast<uint8_t>(cfg_.codec)) {
+      link_down(lk, "protocol error in data stream", false);
+      break;
+    }
+    if (!io_read(lk.fd, lk.recv_pin + 8, SA_ + P_)) {
+      link_down(lk, "payload read failed", true);
+      break;
+    }
+    std::memcpy(scales.data(), lk.recv_pin + 8, 4 * T_);
+    float maxs = 0.f;
+    for (float s : scales) maxs = std::max(maxs, s);
+    lk.rounds_recv++;
+    lk.bytes_recv += 8 + SA_ + P_;
+    lk.last_scale_recv.store(maxs);
+    push_scale(false, maxs);
+    if (maxs == 0.0f) continue;
+    try {
+      apply_packet(lk, scales.data());
+    } catch (const std::exception& e) {
+      link_down(lk, std::string("apply failed: ") + e.what(), false);
+      break;
+    }
+    // wake the other links: they now have fresh residual to forward
+    for (int i = 0; i < 3; ++i) {
+      if (i == lk.idx) continue;
+      Link& o = links_[i];
+      if (o.state.load() == L_ACTIVE) {
+        std::lock_guard<std::mutex> g(o.m);
+        o.dirty = true;
+        o.cv.notify_all();
+      }
+    }
+  }
+}
+
+void Engine::link_down(Link& lk, const std::string& why, bool remote) {
+  int expected = L_ACTIVE;
+  if (!lk.state.compare_exchange_strong(expected, L_DEAD)) return;
+  lk.error = why;
+  if (!remote)
+    set_error("link " + std::to_string(lk.idx) + " (" + lk.peer_desc +
+              ") down: " + why);
+  lk.abort.store(true);
+  if (lk.rccl_link) rccl_abort(static_cast<RcclLink*>(lk.rccl_link));
+  if (lk.fd >= 0) ::shutdown(lk.fd, SHUT_RDWR);
+  lk.cv.notify_all();
+}
+
+// --------------------------------------------------------------- user API
+
+void Engine::notify_all_dirty() {
+  for (auto& lk : links_) {
+    if (lk.state.load() != L_ACTIVE) continue;
+    std::lock_guard<std::mutex> g(lk.m);
+    lk.dirty = true;
+    lk.cv.notify_all();
+  }
+}
+
+void Engine::notify_dirty() { notify_all_dirty(); }
+
+void Engine::add_from(uintptr_t src, int64_t n, uintptr_t stream) {
+  if (n != n_) throw std::runtime_error("add_from: size mismatch");
+  const float* s = reinterpret_cast<const float*>(src);
+  float* d[3];
+  for (int i = 0; i < 3; ++i)
+    d[i] = links_[i].provisioned ? links_[i].delta : nullptr;
+  if (gpu()) {
+    hip_add_scatter(s, n_, 1.0f, values_, d[0], d[1], d[2],
+                    reinterpret_cast<hipStream_t>(stream));
+  } else {
+    float* dsts[4];
+    int nd = 0;
+    dsts[nd++] = values_;
+    for (int i = 0; i < 3; ++i)
+      if (d[i]) dsts[nd++] = d[i];
+    cpu_add_scatter(s, n_, dsts, nd);
+  }
+  notify_all_dirty();
+}
+
+void Engine::copy_to(uintptr_t dst, int64_t n, uintptr_t stream) {
+  if (n != n_) throw std::runtime_error("copy_to: size mismatch");
+  if (gpu()) {
+    HIP_TRY(hipMemcpyAsync(reinterpret_cast<void*>(dst), values_, n_ * 4,
+                           hipMemcpyDeviceToDevice,
+                           reinterpret_cast<hipStream_t>(stream)));
+  } else {
+    float* d = reinterpret_cast<float*>(dst);
+    for (int64_t i = 0; i < n_; ++i) d[i] = atomic_load_f32(values_ + i);
+  }
+}
+
+void Engine::fused_sgd(uintptr_t mom, uintptr_t grad, double lr,
+                       double momentum, uintptr_t stream) {
+  float* d[3];
+  for (int i = 0; i < 3; ++i)
+    d[i] = links_[i].provisioned ? links_[i].delta : nullptr;
+  if (gpu()) {
+    // values is always a destination; pack it with the link deltas
+    hip_fused_sgd(reinterpret_cast<float*>(mom),
+                  reinterpret_cast<const float*>(grad), static_cast<float>(lr),
+                  static_cast<float>(momentum), n_, values_, d[0], d[1], d[2],
+                  reinterpret_cast<hipStream_t>(stream));
+  } else {
+    float* m = reinterpret_cast<float*>(mom);
+    const float* g = reinterpret_cast<const float*>(grad);
+    for (int64_t i = 0; i < n_; ++i) {
+      float mm = static_cast<float>(momentum) * m[i] + g[i];
+      m[i] = mm;
+      float u = static_cast<float>(-lr) * mm;
+      if (u == 0.0f) continue;
+      atomic_add_f32(values_ + i, u);
+      for (int k = 0; k < 3; ++k)
+        if (d[k]) atomic_add_f32(d[k] + i, u);
+    }
+  }
+  notify_all_dirty();
+}
+
+std::vector<LinkStatsSnap> Engine::link_stats() {
+  std::vector<LinkStatsSnap> out;
+  for (auto& lk : links_) {
+    LinkStatsSnap s{};
+    s.rounds_sent = lk.rounds_sent.load();
+    s.rounds_recv = lk.rounds_recv.load();
+    s.bytes_sent = lk.bytes_sent.load();
+    s.bytes_recv = lk.bytes_recv.load();
+    s.last_scale_sent = lk.last_scale_sent.load();
+    s.last_scale_recv = lk.last_scale_recv.load();
+    s.active = lk.state.load() == L_ACTIVE;
+    s.dead = lk.state.load() == L_DEAD;
+    s.peer = lk.peer_desc;
+    s.rccl = lk.rccl;
+    out.push_back(s);
+  }
+  return out;
+}
+
+void Engine::close() {
+  if (closing_.exchange(true)) {
+    // already closing/closed; make idempotent
+  }
+  // tell peers we are leaving (the reference cannot do this and exit(-1)s,
+  // sharedtensor.c:421-430)
+  for (auto& lk : links_) {
+    if (lk.state.load() == L_ACTIVE && lk.fd >= 0) {
+      PacketHeader bye{};
+      bye.type = PKT_CLOSE;
+      bye.ntensors = static_cast<uint32_t>(T_);
+      std::lock_guard<std::mutex> g(lk.wm);
+      io_write(lk.fd, &bye, 8);
+    }
+  }
+  for (auto& lk : links_) {
+    lk.abort.store(true);
+    if (lk.rccl_link) rccl_abort(static_cast<RcclLink*>(lk.rccl_link));
+    if (lk.fd >= 0) ::shutdown(lk.fd, SHUT_RDWR);
+    lk.cv.notify_all();
+  }
+  if (listen_fd_ >= 0) ::shutdown(listen_fd_, SHUT_RDWR);
+  if (listen_thread_.joinable()) listen_thread_.join();
+  for (auto& lk : links_) {
+    if (lk.t_send.joinable()) lk.t_send.join();
+    if (lk.t_recv.joinable()) lk.t_recv.join();
+    if (lk.t_ctrl.joinable()) lk.t_ctrl.join();
+    if (lk.fd >= 0) ::close(lk.fd), lk.fd = -1;
+    if (lk.rccl_link) {
+      rccl_destroy(static_cast<RcclLink*>(lk.rccl_link));
+      lk.rccl_link = nullptr;
+    }
+    lk.state.store(L_DEAD);
+  }
+  if (listen_fd_ >= 0) ::close(listen_fd_), listen_fd_ = -1;
+  free_gpu();
+}
+
+}  // namespace shamd

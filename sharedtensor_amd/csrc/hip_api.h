@@ -1,0 +1,59 @@
+// Host-callable launchers for the CDNA4 codec kernels (hip_kernels.hip).
+//
+// Element-space contract (shared with codec_cpu.cpp / ops/oracle.py):
+//   * values/delta buffers are the UNPADDED concatenation of the table's
+//     tensors (offs[t] element starts, offs[T] = n).
+//   * the packed payload lives in PADDED element space: tensor t occupies
+//     padded elements [poffs[t], poffs[t+1]), poffs multiples of 64, so a
+//     64-lane wavefront never straddles a tensor and __ballot packs one
+//     uint64 word per wave (LSB-first layout identical to the reference's
+//     byte stream, sharedtensor.c:166-177).
+//   * scales: fp32[T] at the head of the message buffer; payload starts at
+//     align8(4*T).
+#pragma once
+
+#include <hip/hip_runtime.h>
+
+#include "common.h"
+
+namespace shamd {
+
+inline int64_t align8(int64_t x) { return (x + 7) & ~int64_t(7); }
+
+struct DevTable {
+  int64_t* offs = nullptr;    // device, T+1 unpadded element offsets
+  int64_t* poffs = nullptr;   // device, T+1 padded element offsets
+  int T = 0;
+  int64_t n = 0;              // offs[T]
+  int64_t pe = 0;             // poffs[T]
+};
+
+// reduce_buf: device scratch, >= 8*T bytes (double sumsq for 1bit,
+// fp32 absmax for fp8/int4). scales_out: device fp32[T].
+void hip_reduce_scales(Codec c, const float* delta, const DevTable& tb,
+                       void* reduce_buf, float* scales_out, int sample_stride,
+                       hipStream_t s);
+
+// delta is debited in place (atomic, lossless vs concurrent adds);
+// payload receives the packed bytes for the whole padded space.
+void hip_quantize(Codec c, float* delta, const DevTable& tb,
+                  const float* scales_dev, uint8_t* payload, hipStream_t s);
+
+// Decode payload and accumulate into up to 4 destinations (values plus the
+// other links' delta buffers, sharedtensor.c:106-127).
+void hip_apply(Codec c, const uint8_t* payload, const DevTable& tb,
+               const float* scales_dev, float* d0, float* d1, float* d2,
+               float* d3, hipStream_t s);
+
+// dst_i += alpha * src over the flat unpadded space (addFromInternal,
+// sharedtensor.c:334-344; alpha=-1 implements the snapshot debit).
+void hip_add_scatter(const float* src, int64_t n, float alpha, float* d0,
+                     float* d1, float* d2, float* d3, hipStream_t s);
+
+// Fused SGD-momentum update feeding the shared tensor: m = mu*m + g;
+// u = -lr*m; {values, link deltas} += u.  One pass over HBM instead of four.
+void hip_fused_sgd(float* mom, const float* grad, float lr, float momentum,
+                   int64_t n, float* d0, float* d1, float* d2, float* d3,
+                   hipStream_t s);
+
+}  // namespace shamd

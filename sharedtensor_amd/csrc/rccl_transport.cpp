@@ -1,0 +1,136 @@
+#include "rccl_transport.h"
+
+#include <rccl/rccl.h>
+
+#include <chrono>
+#include <cstring>
+#include <stdexcept>
+#include <string>
+#include <thread>
+
+namespace shamd {
+
+static_assert(sizeof(ncclUniqueId) == RCCL_ID_BYTES, "ncclUniqueId size");
+
+#define NCCL_TRY(expr)                                                   \
+  do {                                                                   \
+    ncclResult_t _r = (expr);                                            \
+    if (_r != ncclSuccess && _r != ncclInProgress)                       \
+      throw std::runtime_error(std::string("RCCL error: ") +             \
+                               ncclGetErrorString(_r));                  \
+  } while (0)
+
+struct RcclLink {
+  ncclComm_t out = nullptr;  // communicator for this side's sends
+  ncclComm_t in = nullptr;   // communicator for this side's recvs
+  int peer = 0;              // peer rank within each 2-rank comm
+  std::atomic<bool> aborted{false};
+};
+
+void rccl_make_ids(uint8_t ids[2 * RCCL_ID_BYTES]) {
+  ncclUniqueId a, b;
+  NCCL_TRY(ncclGetUniqueId(&a));
+  NCCL_TRY(ncclGetUniqueId(&b));
+  std::memcpy(ids, &a, RCCL_ID_BYTES);
+  std::memcpy(ids + RCCL_ID_BYTES, &b, RCCL_ID_BYTES);
+}
+
+static void wait_comm(ncclComm_t c, double timeout_s, const char* what) {
+  auto deadline = std::chrono::steady_clock::now() +
+                  std::chrono::duration_cast<std::chrono::steady_clock::duration>(
+                      std::chrono::duration<double>(timeout_s));
+  for (;;) {
+    ncclResult_t st;
+    NCCL_TRY(ncclCommGetAsyncError(c, &st));
+    if (st == ncclSuccess) return;
+    if (st != ncclInProgress)
+      throw std::runtime_error(std::string("RCCL ") + what + " failed: " +
+                               ncclGetErrorString(st));
+    if (std::chrono::steady_clock::now() > deadline) {
+      ncclCommAbort(c);
+      throw std::runtime_error(std::string("RCCL ") + what + " timed out");
+    }
+    std::this_thread::sleep_for(std::chrono::milliseconds(2));
+  }
+}
+
+RcclLink* rccl_link_create(int device, const uint8_t ids[2 * RCCL_ID_BYTES],
+                           bool is_parent, double timeout_s) {
+  if (hipSetDevice(device) != hipSuccess)
+    throw std::runtime_error("hipSetDevice failed in rccl_link_create");
+  ncclUniqueId ida, idb;
+  std::memcpy(&ida, ids, RCCL_ID_BYTES);
+  std::memcpy(&idb, ids + RCCL_ID_BYTES, RCCL_ID_BYTES);
+  int my_rank = is_parent ? 0 : 1;
+
+  auto* l = new RcclLink();
+  l->peer = 1 - my_rank;
+  ncclConfig_t cfg = NCCL_CONFIG_INITIALIZER;
+  cfg.blocking = 0;  // poll with timeout so a dead peer cannot hang us
+  try {
+    // direction A (ids[0]): parent -> child payloads; parent sends on it.
+    // direction B (ids[1]): child -> parent payloads.
+    ncclComm_t ca = nullptr, cb = nullptr;
+    NCCL_TRY(ncclCommInitRankConfig(&ca, 2, ida, my_rank, &cfg));
+    NCCL_TRY(ncclCommInitRankConfig(&cb, 2, idb, my_rank, &cfg));
+    wait_comm(ca, timeout_s, "comm A init");
+    wait_comm(cb, timeout_s, "comm B init");
+    l->out = is_parent ? ca : cb;
+    l->in = is_parent ? cb : ca;
+  } catch (...) {
+    delete l;
+    throw;
+  }
+  return l;
+}
+
+static bool wait_stream(RcclLink* l, ncclComm_t c, hipStream_t s,
+                        const std::atomic<bool>& abort) {
+  for (;;) {
+    hipError_t e = hipStreamQuery(s);
+    if (e == hipSuccess) return true;
+    if (e != hipErrorNotReady) return false;
+    ncclResult_t st;
+    if (ncclCommGetAsyncError(c, &st) != ncclSuccess || (st != ncclSuccess && st != ncclInProgress))
+      return false;
+    if (abort.load() || l->aborted.load()) {
+      ncclCommAbort(c);
+      l->aborted.store(true);
+      hipStreamSynchronize(s);
+      return false;
+    }
+    std::this_thread::sleep_for(std::chrono::microseconds(20));
+  }
+}
+
+bool rccl_send(RcclLink* l, const void* buf, size_t bytes, hipStream_t stream,
+               const std::atomic<bool>& abort) {
+  if (l->aborted.load()) return false;
+  if (ncclSend(buf, bytes, ncclChar, l->peer, l->out, stream) != ncclSuccess)
+    return false;
+  return wait_stream(l, l->out, stream, abort);
+}
+
+bool rccl_recv(RcclLink* l, void* buf, size_t bytes, hipStream_t stream,
+               const std::atomic<bool>& abort) {
+  if (l->aborted.load()) return false;
+  if (ncclRecv(buf, bytes, ncclChar, l->peer, l->in, stream) != ncclSuccess)
+    return false;
+  return wait_stream(l, l->in, stream, abort);
+}
+
+void rccl_abort(RcclLink* l) {
+  if (!l || l->aborted.exchange(true)) return;
+  if (l->out) ncclCommAbort(l->out);
+  if (l->in) ncclCommAbort(l->in);
+}
+
+void rccl_destroy(RcclLink* l) {
+  if (!l) return;
+  rccl_abort(l);
+  if (l->out) ncclCommDestroy(l->out);
+  if (l->in) ncclCommDestroy(l->in);
+  delete l;
+}
+
+}  // namespace shamd

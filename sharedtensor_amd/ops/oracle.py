@@ -75,10 +75,17 @@ def rms_scale(delta: torch.Tensor) -> float:
 
 # ---------------------------------------------------------------- 1-bit codec
 
+def pad64(n: int) -> int:
+    """Padded element count: each tensor's payload region covers a multiple
+    of 64 elements so a 64-lane CDNA4 wavefront never straddles a tensor and
+    __ballot packs exactly one uint64 word per wave."""
+    return (n + 63) // 64 * 64
+
+
 def words_1bit(n: int) -> int:
     """Payload size in uint64 words (8-byte aligned; superset of the
     reference's ceil(n/8) bytes — the first ceil(n/8) bytes are identical)."""
-    return (n + 63) // 64
+    return pad64(n) // 64
 
 
 def encode_1bit(delta: torch.Tensor, scale: float = None) -> Tuple[float, bytes, torch.Tensor]:
@@ -136,7 +143,7 @@ def encode_fp8(delta: torch.Tensor, scale: float = None) -> Tuple[float, bytes, 
     if scale is None:
         scale = fp8_scale(d)
     scale = np.float32(scale).item()
-    pad = (-n) % 8
+    pad = pad64(n) - n
     if scale == 0.0:
         return 0.0, b"\x00" * (n + pad), d
     q = torch.clamp(d / scale, -FP8_MAX, FP8_MAX).to(torch.float8_e4m3fn)
@@ -174,7 +181,7 @@ def encode_int4(delta: torch.Tensor, scale: float = None) -> Tuple[float, bytes,
         scale = int4_scale(d)
     scale = np.float32(scale).item()
     nbytes = (n + 1) // 2
-    pad = (-nbytes) % 8
+    pad = pad64(n) // 2 - nbytes
     if scale == 0.0:
         return 0.0, b"\x00" * (nbytes + pad), d
     q = torch.clamp(torch.round(d / scale), -7, 7).to(torch.int8)
@@ -207,12 +214,11 @@ def decode_int4(payload: bytes, scale: float, n: int) -> torch.Tensor:
 
 def payload_bytes(codec: int, n: int) -> int:
     if codec == CODEC_1BIT:
-        return words_1bit(n) * 8
+        return pad64(n) // 8
     if codec == CODEC_FP8:
-        return n + ((-n) % 8)
+        return pad64(n)
     if codec == CODEC_INT4:
-        nb = (n + 1) // 2
-        return nb + ((-nb) % 8)
+        return pad64(n) // 2
     raise ValueError(f"unknown codec {codec}")
 
 

@@ -3,7 +3,15 @@ import sys
 
 import pytest
 
-sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+sys.path.insert(0, REPO)
+
+# build the native extension if missing/stale (no-op when up to date)
+try:
+    import build as _build_mod
+    _build_mod.build(verbose=False)
+except Exception as e:  # pragma: no cover
+    print(f"[conftest] native build failed: {e}", file=sys.stderr)
 
 
 def pytest_configure(config):
