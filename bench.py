@@ -1,0 +1,268 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: async-DP GPT-2-small training over the shared-tensor
+engine (BASELINE.json config 3), plus a param-sync bandwidth mode.
+
+Contract (driver): `python bench.py --gpus N --steps K --warmup W` runs one
+rank per GPU (launched via torch.distributed.run for N>1; reads RANK /
+WORLD_SIZE / LOCAL_RANK from the env), does W untimed warmup steps, times
+exactly K steps bracketed by barrier + torch.cuda.synchronize on both sides,
+takes the MAX step time over ranks, and rank 0 prints ONE JSON line.
+
+The metric is the whole-job aggregate tokens/s; config reports the engine's
+param-sync wire GB/s and p50 staleness (the per-round scale: every packet
+moves each element by exactly +-scale).  torch.distributed (gloo) is used
+ONLY for barriers and the max-reduce of the measured time — the parameter
+sync data plane is this framework's engine (TCP tree / RCCL over xGMI).
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+
+def log(msg):
+    r = os.environ.get("RANK", "0")
+    print(f"[bench r{r}] {msg}", file=sys.stderr, flush=True)
+
+
+def dist_setup(world):
+    import torch.distributed as dist
+    if world > 1 and not dist.is_initialized():
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29531")
+        dist.init_process_group("gloo", rank=int(os.environ.get("RANK", 0)),
+                                world_size=world)
+    return dist if world > 1 else None
+
+
+def barrier(dist):
+    if dist is not None:
+        dist.barrier()
+
+
+def max_over_ranks(dist, x: float) -> float:
+    if dist is None:
+        return x
+    t = torch.tensor([x], dtype=torch.float64)
+    dist.all_reduce(t, op=dist.ReduceOp.MAX)
+    return t.item()
+
+
+def sum_over_ranks(dist, x: float) -> float:
+    if dist is None:
+        return x
+    t = torch.tensor([x], dtype=torch.float64)
+    dist.all_reduce(t, op=dist.ReduceOp.SUM)
+    return t.item()
+
+
+def run_train(args, rank, world, device):
+    from sharedtensor_amd.models.gpt2 import GPT2, GPT2Config
+    from sharedtensor_amd.parallel.async_dp import AsyncDPTrainer
+
+    cfg = GPT2Config.tiny() if args.model == "tiny" else GPT2Config.small()
+    if args.seq:
+        cfg.block_size = min(cfg.block_size, args.seq) if args.model == "tiny" else args.seq
+    torch.manual_seed(1234)  # same random init on every rank
+    model = GPT2(cfg).to(device)
+    log(f"model {args.model}: {model.num_params()/1e6:.1f}M params, device {device}")
+
+    trainer = AsyncDPTrainer(
+        model, host="127.0.0.1",
+        port_base=int(os.environ.get("SHTENS_PORT_BASE", 52000)),
+        rank=rank, world=world, lr=args.lr, momentum=0.9,
+        amp_dtype=torch.bfloat16 if device.type == "cuda" else None,
+        codec=args.codec, use_rccl=not args.no_rccl,
+        snapshot_join=True)
+
+    B, T = args.batch, cfg.block_size
+    gen = torch.Generator(device="cpu").manual_seed(42 + rank)
+    batches = [torch.randint(0, cfg.vocab_size, (B, T + 1), generator=gen).to(device)
+               for _ in range(4)]
+
+    dist = dist_setup(world)
+
+    def one_step(i):
+        b = batches[i % len(batches)]
+        loss = trainer.step(b[:, :-1], b[:, 1:])
+        return loss
+
+    log(f"warmup {args.warmup} steps")
+    for i in range(args.warmup):
+        one_step(i)
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+    barrier(dist)
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+    s0 = trainer.stats()
+    t0 = time.perf_counter()
+    last_loss = None
+    for i in range(args.steps):
+        last_loss = one_step(args.warmup + i)
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+    t1 = time.perf_counter()
+    barrier(dist)
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+    s1 = trainer.stats()
+    dt = max_over_ranks(dist, t1 - t0)
+
+    wire_bytes = sum_over_ranks(
+        dist, float((s1["bytes_sent"] - s0["bytes_sent"]) +
+                    (s1["bytes_recv"] - s0["bytes_recv"])))
+    rounds = sum_over_ranks(
+        dist, float((s1["rounds_sent"] - s0["rounds_sent"])))
+    n_params = trainer.shared.n
+    tokens = float(B * T * args.steps * world)
+    value = tokens / dt
+    result = {
+        "metric": "async-DP tokens/sec GPT-2-small",
+        "value": round(value, 1),
+        "unit": "tokens/s",
+        "n_gpus": world,
+        "steps": args.steps,
+        "warmup": args.warmup,
+        "ms_per_step": round(dt / args.steps * 1e3, 3),
+        "higher_is_better": True,
+        "scaling": "weak",
+        "vs_baseline": None,  # reference publishes no numbers
+        "dtype": "bf16" if device.type == "cuda" else "fp32",
+        "data": "synthetic",
+        "config": {
+            "model": "gpt2-small" if args.model == "small" else args.model,
+            "global_batch": B * world,
+            "seq_len": T,
+            "parallelism": f"async-dp{world}",
+            "codec": args.codec,
+            "n_params": n_params,
+            "loss": round(float(last_loss), 4) if last_loss is not None else None,
+            "paramsync_wire_gbps": round(wire_bytes / dt / 1e9, 3),
+            "paramsync_logical_gbps": round(rounds * n_params * 4 / dt / 1e9, 3),
+            "sync_rounds_per_s": round(rounds / dt, 1),
+            "staleness_p50": s1["staleness_p50"],
+            "staleness_p90": s1["staleness_p90"],
+        },
+    }
+    trainer.close()
+    if dist is not None:
+        dist.barrier()
+        dist.destroy_process_group()
+    if rank == 0:
+        print(json.dumps(result), flush=True)
+
+
+def run_paramsync(args, rank, world, device):
+    """Secondary mode: raw param-sync bandwidth on a flat tensor
+    (BASELINE configs 2 and 5)."""
+    from sharedtensor_amd.engine import _SharedBase
+    from sharedtensor_amd.parallel.async_dp import tree_children, tree_parent
+
+    n = args.numel
+    port_base = int(os.environ.get("SHTENS_PORT_BASE", 52000))
+    nchild = len(tree_children(rank, world))
+    sh = _SharedBase(
+        "127.0.0.1", port_base, [n], device=device, codec=args.codec,
+        use_rccl=not args.no_rccl, expected_children=nchild if world > 1 else 0,
+        provision_up=rank > 0,
+        explicit_parent=f"127.0.0.1:{port_base + tree_parent(rank)}" if rank else "",
+        listen_port=port_base + rank if world > 1 else 0)
+    sh._start()
+    dist = dist_setup(world)
+    delta = torch.randn(n, dtype=torch.float32, device=device) * 0.01
+
+    def one_step():
+        sh._add_flat(delta)
+        if device.type == "cuda":
+            torch.cuda.synchronize()
+        time.sleep(args.interval)
+
+    for _ in range(args.warmup):
+        one_step()
+    barrier(dist)
+    s0 = sh.stats()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        one_step()
+    # let the engine drain for a fixed window so we measure sync, not add
+    time.sleep(1.0)
+    t1 = time.perf_counter()
+    barrier(dist)
+    s1 = sh.stats()
+    dt = max_over_ranks(dist, t1 - t0)
+    wire = sum_over_ranks(dist, float(s1["bytes_sent"] - s0["bytes_sent"] +
+                                      s1["bytes_recv"] - s0["bytes_recv"]))
+    rounds = sum_over_ranks(dist, float(s1["rounds_sent"] - s0["rounds_sent"]))
+    result = {
+        "metric": "param-sync GB/s (flat tensor)",
+        "value": round(rounds * n * 4 / dt / 1e9, 3),
+        "unit": "GB/s logical",
+        "n_gpus": world,
+        "steps": args.steps,
+        "warmup": args.warmup,
+        "ms_per_step": round(dt / args.steps * 1e3, 3),
+        "higher_is_better": True,
+        "scaling": "weak",
+        "vs_baseline": None,
+        "dtype": "fp32",
+        "data": "synthetic",
+        "config": {
+            "numel": n, "codec": args.codec,
+            "wire_gbps": round(wire / dt / 1e9, 3),
+            "staleness_p50": s1["staleness_p50"],
+        },
+    }
+    sh.close()
+    if dist is not None:
+        dist.barrier()
+        dist.destroy_process_group()
+    if rank == 0:
+        print(json.dumps(result), flush=True)
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--batch", type=int, default=8)
+    ap.add_argument("--seq", type=int, default=1024)
+    ap.add_argument("--lr", type=float, default=0.01)
+    ap.add_argument("--codec", choices=["1bit", "fp8", "int4"], default="1bit")
+    ap.add_argument("--model", choices=["small", "tiny"], default="small")
+    ap.add_argument("--mode", choices=["train", "paramsync"], default="train")
+    ap.add_argument("--numel", type=int, default=268_435_456)  # 1 GB fp32
+    ap.add_argument("--interval", type=float, default=0.01)
+    ap.add_argument("--no-rccl", action="store_true")
+    ap.add_argument("--device", default="auto")
+    args = ap.parse_args()
+
+    rank = int(os.environ.get("RANK", 0))
+    world = int(os.environ.get("WORLD_SIZE", args.gpus))
+    local = int(os.environ.get("LOCAL_RANK", rank))
+    if args.device == "auto":
+        device = torch.device(f"cuda:{local}") if torch.cuda.is_available() \
+            else torch.device("cpu")
+    else:
+        device = torch.device(args.device)
+    if device.type == "cuda":
+        torch.cuda.set_device(device)
+    if device.type == "cpu" and args.model == "small" and args.mode == "train":
+        log("no GPU: downshifting to tiny model for a functional run")
+        args.model = "tiny"
+        args.batch = min(args.batch, 2)
+
+    if args.mode == "train":
+        run_train(args, rank, world, device)
+    else:
+        run_paramsync(args, rank, world, device)
+
+
+if __name__ == "__main__":
+    main()

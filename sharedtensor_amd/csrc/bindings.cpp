@@ -39,6 +39,87 @@ double py_cpu_scale(int codec, uintptr_t delta, int64_t n, int stride) {
                            reinterpret_cast<float*>(delta), n, stride);
 }
 
+// Direct handle on the CDNA4 codec kernels, for numerics tests and ad-hoc
+// use (tensors pass as data_ptr integers; caller owns all buffers).
+struct DevCodec {
+  DevTable tb{};
+  Codec c;
+  int device;
+  void* reduce = nullptr;
+
+  DevCodec(int codec, std::vector<int64_t> sizes, int dev)
+      : c(static_cast<Codec>(codec)), device(dev) {
+    std::vector<int64_t> offs(sizes.size() + 1), poffs(sizes.size() + 1);
+    offs[0] = poffs[0] = 0;
+    for (size_t t = 0; t < sizes.size(); ++t) {
+      offs[t + 1] = offs[t] + sizes[t];
+      poffs[t + 1] = poffs[t] + pad64(sizes[t]);
+    }
+    if (hipSetDevice(device) != hipSuccess)
+      throw std::runtime_error("hipSetDevice failed");
+    int T = static_cast<int>(sizes.size());
+    if (hipMalloc(&tb.offs, sizeof(int64_t) * (T + 1) * 2) != hipSuccess)
+      throw std::runtime_error("hipMalloc failed");
+    tb.poffs = tb.offs + (T + 1);
+    hipMemcpy(tb.offs, offs.data(), sizeof(int64_t) * (T + 1),
+              hipMemcpyHostToDevice);
+    hipMemcpy(tb.poffs, poffs.data(), sizeof(int64_t) * (T + 1),
+              hipMemcpyHostToDevice);
+    tb.T = T;
+    tb.n = offs[T];
+    tb.pe = poffs[T];
+    if (hipMalloc(&reduce, 8 * T) != hipSuccess)
+      throw std::runtime_error("hipMalloc failed");
+  }
+  ~DevCodec() {
+    if (tb.offs) hipFree(tb.offs);
+    if (reduce) hipFree(reduce);
+  }
+  void reduce_scales(uintptr_t delta, uintptr_t scales_dev, int stride,
+                     uintptr_t stream) {
+    hip_reduce_scales(c, reinterpret_cast<const float*>(delta), tb, reduce,
+                      reinterpret_cast<float*>(scales_dev), stride,
+                      reinterpret_cast<hipStream_t>(stream));
+  }
+  void quantize(uintptr_t delta, uintptr_t scales_dev, uintptr_t payload,
+                uintptr_t stream) {
+    hip_quantize(c, reinterpret_cast<float*>(delta), tb,
+                 reinterpret_cast<const float*>(scales_dev),
+                 reinterpret_cast<uint8_t*>(payload),
+                 reinterpret_cast<hipStream_t>(stream));
+  }
+  void apply(uintptr_t payload, uintptr_t scales_dev,
+             std::vector<uintptr_t> dsts, uintptr_t stream) {
+    float* d[4] = {nullptr, nullptr, nullptr, nullptr};
+    for (size_t i = 0; i < dsts.size() && i < 4; ++i)
+      d[i] = reinterpret_cast<float*>(dsts[i]);
+    hip_apply(c, reinterpret_cast<const uint8_t*>(payload), tb,
+              reinterpret_cast<const float*>(scales_dev), d[0], d[1], d[2],
+              d[3], reinterpret_cast<hipStream_t>(stream));
+  }
+};
+
+void py_gpu_add_scatter(uintptr_t src, int64_t n, double alpha,
+                        std::vector<uintptr_t> dsts, uintptr_t stream) {
+  float* d[4] = {nullptr, nullptr, nullptr, nullptr};
+  for (size_t i = 0; i < dsts.size() && i < 4; ++i)
+    d[i] = reinterpret_cast<float*>(dsts[i]);
+  hip_add_scatter(reinterpret_cast<const float*>(src), n,
+                  static_cast<float>(alpha), d[0], d[1], d[2], d[3],
+                  reinterpret_cast<hipStream_t>(stream));
+}
+
+void py_gpu_fused_sgd(uintptr_t mom, uintptr_t grad, double lr, double mu,
+                      int64_t n, std::vector<uintptr_t> dsts, uintptr_t stream) {
+  float* d[4] = {nullptr, nullptr, nullptr, nullptr};
+  for (size_t i = 0; i < dsts.size() && i < 4; ++i)
+    d[i] = reinterpret_cast<float*>(dsts[i]);
+  hip_fused_sgd(reinterpret_cast<float*>(mom),
+                reinterpret_cast<const float*>(grad), static_cast<float>(lr),
+                static_cast<float>(mu), n, d[0], d[1], d[2], d[3],
+                reinterpret_cast<hipStream_t>(stream));
+}
+
 }  // namespace
 
 PYBIND11_MODULE(_core, m) {
@@ -60,6 +141,7 @@ PYBIND11_MODULE(_core, m) {
       .def_readwrite("expected_children", &Config::expected_children)
       .def_readwrite("sizes", &Config::sizes)
       .def_readwrite("explicit_parent", &Config::explicit_parent)
+      .def_readwrite("listen_port", &Config::listen_port)
       .def_readwrite("join_timeout_s", &Config::join_timeout_s)
       .def_readwrite("rms_sample_stride", &Config::rms_sample_stride);
 
@@ -99,6 +181,14 @@ PYBIND11_MODULE(_core, m) {
         }
         return out;
       });
+
+  py::class_<DevCodec>(m, "DevCodec")
+      .def(py::init<int, std::vector<int64_t>, int>())
+      .def("reduce_scales", &DevCodec::reduce_scales)
+      .def("quantize", &DevCodec::quantize)
+      .def("apply", &DevCodec::apply);
+  m.def("gpu_add_scatter", &py_gpu_add_scatter);
+  m.def("gpu_fused_sgd", &py_gpu_fused_sgd);
 
   m.def("msg_bytes", &Engine::msg_bytes);
   m.def("scales_area", &Engine::scales_area);

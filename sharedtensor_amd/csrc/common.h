@@ -56,6 +56,9 @@ struct Config {
                                // size 1 = classic flat tensor)
   std::string explicit_parent;  // "ip:port" — skip the discovery walk and
                                 // connect straight to this node ("" = walk)
+  int listen_port = 0;  // >0: bind the listener to this fixed port (explicit
+                        // topologies); 0: reference behaviour — the local
+                        // port of the up-connection
   double join_timeout_s = 60.0;
   int rms_sample_stride = 1;  // >1: subsample the RMS reduction (scale is a
                               // heuristic; stride k cuts its HBM traffic k×)

@@ -368,6 +368,8 @@ void Engine::handshake_as_child(int fd) {
   sockaddr_in self{};
   socklen_t slen = sizeof(self);
   getsockname(fd, reinterpret_cast<sockaddr*>(&self), &slen);
+  if (cfg_.listen_port > 0)
+    self.sin_port = htons(static_cast<uint16_t>(cfg_.listen_port));
   bind_listen(self);
 
   if (ah.flags & ACC_RCCL) {
@@ -716,7 +718,7 @@ void Engine::send_loop(Link& lk) {
         ping.ntensors = static_cast<uint32_t>(T_);
         std::lock_guard<std::mutex> g(lk.wm);
         if (!io_write(lk.fd, &ping, 8)) {
-          link_down(lk, "keepalive write failed", false);
+          link_down(lk, "keepalive write failed", true);
           break;
         }
         last_send = now;
@@ -759,7 +761,7 @@ void Engine::send_loop(Link& lk) {
         break;
       }
     } else if (!send_packet(lk, scales.data())) {
-      link_down(lk, "data write failed", false);
+      link_down(lk, "data write failed", true);
       break;
     }
     last_send = Clock::now();

@@ -40,7 +40,8 @@ class _SharedBase:
                  reconnect: bool = False, keepalive_s: float = 1.0,
                  bw_limit: float = 0.0, expected_children: int = 2,
                  provision_up: bool = True, explicit_parent: str = "",
-                 join_timeout_s: float = 60.0, rms_sample_stride: int = 1):
+                 listen_port: int = 0, join_timeout_s: float = 60.0,
+                 rms_sample_stride: int = 1):
         if codec not in CODECS:
             raise ValueError(f"codec must be one of {list(CODECS)}")
         self.device = torch.device(device)
@@ -58,6 +59,7 @@ class _SharedBase:
         cfg.expected_children = expected_children
         cfg.sizes = [int(s) for s in sizes]
         cfg.explicit_parent = explicit_parent
+        cfg.listen_port = int(listen_port)
         cfg.join_timeout_s = join_timeout_s
         cfg.rms_sample_stride = rms_sample_stride
         self.codec = codec
