@@ -238,10 +238,17 @@ void Engine::start() {
   listen_thread_ = std::thread([this] { listen_loop(); });
 }
 
-bool Engine::try_connect(const sockaddr_in& addr, int& out_fd) {
+bool Engine::try_connect(const sockaddr_in& addr, int& out_fd,
+                         const sockaddr_in* bind_local) {
   int fd = ::socket(AF_INET, SOCK_STREAM, 0);
   if (fd < 0) return false;
   set_sockopts(fd);
+  if (bind_local &&
+      ::bind(fd, reinterpret_cast<const sockaddr*>(bind_local),
+             sizeof(*bind_local)) < 0) {
+    ::close(fd);
+    return false;
+  }
   if (::connect(fd, reinterpret_cast<const sockaddr*>(&addr), sizeof(addr)) < 0) {
     ::close(fd);
     return false;
@@ -344,7 +351,7 @@ void Engine::become_master() {
                T_ == 1 ? "" : "s");
 }
 
-void Engine::handshake_as_child(int fd) {
+void Engine::handshake_as_child(int fd, bool rejoin) {
   AcceptHello ah{};
   if (!io_read(fd, &ah, sizeof(ah)))
     throw std::runtime_error("parent hung up during handshake");
@@ -362,15 +369,17 @@ void Engine::handshake_as_child(int fd) {
   getpeername(fd, reinterpret_cast<sockaddr*>(&up.peer), &alen);
   up.peer_desc = addr_str(up.peer);
 
-  // bind our listener to the local address of the up socket so our parent's
-  // view of us (getpeername) is also our listen address — the reference's
-  // self-addressing trick (sharedtensor.c:292-316)
-  sockaddr_in self{};
-  socklen_t slen = sizeof(self);
-  getsockname(fd, reinterpret_cast<sockaddr*>(&self), &slen);
-  if (cfg_.listen_port > 0)
-    self.sin_port = htons(static_cast<uint16_t>(cfg_.listen_port));
-  bind_listen(self);
+  if (!rejoin) {
+    // bind our listener to the local address of the up socket so our
+    // parent's view of us (getpeername) is also our listen address — the
+    // reference's self-addressing trick (sharedtensor.c:292-316)
+    sockaddr_in self{};
+    socklen_t slen = sizeof(self);
+    getsockname(fd, reinterpret_cast<sockaddr*>(&self), &slen);
+    if (cfg_.listen_port > 0)
+      self.sin_port = htons(static_cast<uint16_t>(cfg_.listen_port));
+    bind_listen(self);
+  }
 
   if (ah.flags & ACC_RCCL) {
     uint8_t ids[2 * RCCL_ID_BYTES];
@@ -378,10 +387,218 @@ void Engine::handshake_as_child(int fd) {
       throw std::runtime_error("failed to read RCCL ids from parent");
     rccl_upgrade(up, ids, /*is_parent=*/false);
   }
-  if (ah.flags & ACC_SNAPSHOT) recv_snapshot(fd);
+  if (rejoin && (ah.flags & ACC_SNAPSHOT)) {
+    // reconciliation on rejoin: V := S + tmp, where S is the new parent's
+    // snapshot and tmp = our unsent up-residual (preserved in up.delta).
+    // Child slots were demoted/zeroed, so the unconnected-slot invariant
+    // (slot delta == values) is rebuilt by the same adds.
+    zero_buf(values_, n_);
+    recv_snapshot(fd);  // += S into values + child slots
+    float* fwd0 = links_[LK_LEFT].provisioned ? links_[LK_LEFT].delta : nullptr;
+    float* fwd1 = links_[LK_RIGHT].provisioned ? links_[LK_RIGHT].delta : nullptr;
+    if (gpu()) {
+      hip_add_scatter(up.delta, n_, 1.0f, values_, fwd0, fwd1, nullptr,
+                      up.s_recv);
+      HIP_TRY(hipStreamSynchronize(up.s_recv));
+    } else {
+      for (int64_t i = 0; i < n_; ++i) {
+        float v = atomic_load_f32(up.delta + i);
+        if (v == 0.0f) continue;
+        atomic_add_f32(values_ + i, v);
+        if (fwd0) atomic_add_f32(fwd0 + i, v);
+        if (fwd1) atomic_add_f32(fwd1 + i, v);
+      }
+    }
+  } else if (ah.flags & ACC_SNAPSHOT) {
+    recv_snapshot(fd);
+  }
 
   up.state.store(L_ACTIVE);
   spawn_link_threads(up);
+}
+
+void Engine::zero_buf(float* p, int64_t n) {
+  if (gpu()) {
+    HIP_TRY(hipMemset(p, 0, n * 4));
+  } else {
+    std::memset(p, 0, n * 4);
+  }
+}
+
+void Engine::drop_children() {
+  std::lock_guard<std::mutex> g(slots_m_);
+  for (int i : {LK_LEFT, LK_RIGHT}) {
+    Link& lk = links_[i];
+    if (lk.state.load() == L_ACTIVE && lk.fd >= 0) {
+      PacketHeader bye{};
+      bye.type = PKT_CLOSE;
+      bye.ntensors = static_cast<uint32_t>(T_);
+      std::lock_guard<std::mutex> wg(lk.wm);
+      io_write(lk.fd, &bye, 8);
+    }
+    link_down(lk, "demoted for rejoin", true);
+    if (lk.t_send.joinable()) lk.t_send.join();
+    if (lk.t_recv.joinable()) lk.t_recv.join();
+    if (lk.t_ctrl.joinable()) lk.t_ctrl.join();
+    if (lk.fd >= 0) ::close(lk.fd), lk.fd = -1;
+    if (lk.rccl_link) {
+      rccl_destroy(static_cast<RcclLink*>(lk.rccl_link));
+      lk.rccl_link = nullptr;
+      lk.rccl = false;
+    }
+    lk.abort.store(false);
+    lk.error.clear();
+    if (lk.provisioned) zero_buf(lk.delta, n_);
+    lk.state.store(L_FREE);
+  }
+}
+
+bool Engine::failover_master() {
+  // the root is gone: race to take over the rendezvous address; loser
+  // restores its own listener and rejoins the winner
+  sockaddr_in old_addr = listen_addr_;
+  if (listen_fd_ >= 0) ::shutdown(listen_fd_, SHUT_RDWR);
+  if (listen_thread_.joinable()) listen_thread_.join();
+  if (listen_fd_ >= 0) ::close(listen_fd_), listen_fd_ = -1;
+  try {
+    bind_listen(root_addr_);
+    is_master_ = true;
+  } catch (const std::exception&) {
+    try {
+      bind_listen(old_addr);  // reclaim our previous address
+      listen_thread_ = std::thread([this] { listen_loop(); });
+    } catch (const std::exception& e2) {
+      set_error(std::string("failover: lost listener: ") + e2.what());
+    }
+    return false;
+  }
+  listen_thread_ = std::thread([this] { listen_loop(); });
+  std::fprintf(stderr, "[sharedtensor_amd] failover: now master at %s\n",
+               addr_str(root_addr_).c_str());
+  return true;
+}
+
+// Rejoin after losing the up link (the reference's acknowledged missing
+// feature, README.md:33).  Demote-then-rejoin: children are released (they
+// rejoin through the root themselves, which keeps the topology acyclic),
+// then this node re-enters the tree carrying its unsent residual.
+void Engine::reconnect_loop() try {
+  Link& up = links_[LK_UP];
+  if (up.t_send.joinable()) up.t_send.join();
+  if (up.t_recv.joinable()) up.t_recv.join();
+  if (up.t_ctrl.joinable()) up.t_ctrl.join();
+  if (up.fd >= 0) ::close(up.fd), up.fd = -1;
+  if (up.rccl_link) {
+    rccl_destroy(static_cast<RcclLink*>(up.rccl_link));
+    up.rccl_link = nullptr;
+    up.rccl = false;
+  }
+  up.abort.store(false);
+  up.state.store(L_FREE);
+  drop_children();
+
+  bool explicit_mode = !cfg_.explicit_parent.empty();
+  sockaddr_in target = root_addr_;
+  if (explicit_mode) {
+    auto pos = cfg_.explicit_parent.rfind(':');
+    resolve_ipv4(cfg_.explicit_parent.substr(0, pos),
+                 std::stoi(cfg_.explicit_parent.substr(pos + 1)), &target);
+  }
+  auto deadline = Clock::now() + std::chrono::duration_cast<Clock::duration>(
+      std::chrono::duration<double>(cfg_.join_timeout_s));
+  int hops = 0;
+  while (!closing_) {
+    if (Clock::now() > deadline) {
+      set_error("rejoin timed out");
+      break;
+    }
+    int fd = -1;
+    sockaddr_in local = listen_addr_;
+    if (!try_connect(target, fd, &local)) {
+      if (!explicit_mode && hops == 0 && !closing_) {
+        if (failover_master()) {
+          // restore the unconnected-slot invariant (slot delta == values,
+          // sharedtensor.c:379-381 semantics): future children must receive
+          // the full inherited state
+          float* fwd0 = links_[LK_LEFT].provisioned ? links_[LK_LEFT].delta : nullptr;
+          float* fwd1 = links_[LK_RIGHT].provisioned ? links_[LK_RIGHT].delta : nullptr;
+          if (fwd0 || fwd1) {
+            if (gpu()) {
+              HIP_TRY(hipSetDevice(cfg_.device));
+              hip_add_scatter(values_, n_, 1.0f, fwd0, fwd1, nullptr, nullptr,
+                              nullptr);
+              HIP_TRY(hipStreamSynchronize(nullptr));
+            } else {
+              for (int64_t i = 0; i < n_; ++i) {
+                float v = atomic_load_f32(values_ + i);
+                if (v == 0.0f) continue;
+                if (fwd0) atomic_add_f32(fwd0 + i, v);
+                if (fwd1) atomic_add_f32(fwd1 + i, v);
+              }
+            }
+          }
+          break;
+        }
+      }
+      std::this_thread::sleep_for(std::chrono::milliseconds(250));
+      target = explicit_mode ? target : root_addr_;
+      hops = 0;
+      continue;
+    }
+    Hello h{};
+    h.magic = MAGIC;
+    h.version = PROTO_VERSION;
+    h.flags = static_cast<uint16_t>((gpu() ? HELLO_HAS_GPU : 0) |
+                                    (gpu() && cfg_.use_rccl ? HELLO_WANT_RCCL : 0));
+    h.n = static_cast<uint64_t>(n_);
+    h.ntensors = static_cast<uint32_t>(T_);
+    h.codec = static_cast<uint32_t>(cfg_.codec);
+    h.hostid = hostid_;
+    h.device = cfg_.device;
+    uint8_t reply = 0;
+    if (!io_write(fd, &h, sizeof(h)) || !io_read(fd, &reply, 1)) {
+      ::close(fd);
+      std::this_thread::sleep_for(std::chrono::milliseconds(100));
+      target = explicit_mode ? target : root_addr_;
+      hops = 0;
+      continue;
+    }
+    if (reply == 'N') {
+      uint8_t buf[6];
+      if (io_read(fd, buf, 6)) {
+        std::memcpy(&target.sin_addr.s_addr, buf, 4);
+        std::memcpy(&target.sin_port, buf + 4, 2);
+        hops++;
+      } else {
+        target = explicit_mode ? target : root_addr_;
+        hops = 0;
+      }
+      ::close(fd);
+      continue;
+    }
+    if (reply != 'Y') {
+      ::close(fd);
+      std::this_thread::sleep_for(std::chrono::milliseconds(250));
+      continue;
+    }
+    try {
+      handshake_as_child(fd, /*rejoin=*/true);
+      reconnects_++;
+      set_error("");
+      break;
+    } catch (const std::exception& e) {
+      ::close(fd);
+      up.fd = -1;
+      set_error(std::string("rejoin handshake failed: ") + e.what());
+      std::this_thread::sleep_for(std::chrono::milliseconds(250));
+      target = explicit_mode ? target : root_addr_;
+      hops = 0;
+    }
+  }
+  reconnecting_.store(false);
+} catch (const std::exception& e) {
+  set_error(std::string("reconnect failed: ") + e.what());
+  reconnecting_.store(false);
 }
 
 void Engine::bind_listen(const sockaddr_in& addr) {
@@ -404,6 +621,7 @@ void Engine::bind_listen(const sockaddr_in& addr) {
   socklen_t glen = sizeof(got);
   getsockname(listen_fd_, reinterpret_cast<sockaddr*>(&got), &glen);
   listen_port_ = ntohs(got.sin_port);
+  listen_addr_ = got;
 }
 
 // ----------------------------------------------------------------- listen
@@ -897,6 +1115,13 @@ void Engine::link_down(Link& lk, const std::string& why, bool remote) {
   if (lk.rccl_link) rccl_abort(static_cast<RcclLink*>(lk.rccl_link));
   if (lk.fd >= 0) ::shutdown(lk.fd, SHUT_RDWR);
   lk.cv.notify_all();
+  if (lk.idx == LK_UP && cfg_.reconnect && !closing_) {
+    bool expected = false;
+    if (reconnecting_.compare_exchange_strong(expected, true)) {
+      if (reconnect_thread_.joinable()) reconnect_thread_.join();
+      reconnect_thread_ = std::thread([this] { reconnect_loop(); });
+    }
+  }
 }
 
 // --------------------------------------------------------------- user API
@@ -1011,6 +1236,7 @@ void Engine::close() {
     if (lk.fd >= 0) ::shutdown(lk.fd, SHUT_RDWR);
     lk.cv.notify_all();
   }
+  if (reconnect_thread_.joinable()) reconnect_thread_.join();
   if (listen_fd_ >= 0) ::shutdown(listen_fd_, SHUT_RDWR);
   if (listen_thread_.joinable()) listen_thread_.join();
   for (auto& lk : links_) {

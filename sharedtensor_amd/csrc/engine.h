@@ -93,7 +93,12 @@ class Engine {
   Link links_[3];
   int listen_fd_ = -1;
   int listen_port_ = 0;
+  sockaddr_in listen_addr_{};
+  std::mutex slots_m_;  // child-slot assignment vs reconnect demotion
   std::thread listen_thread_;
+  std::thread reconnect_thread_;
+  std::atomic<bool> reconnecting_{false};
+  std::atomic<uint64_t> reconnects_{0};
   std::atomic<bool> closing_{false};
   bool started_ = false;
   bool is_master_ = false;
@@ -111,8 +116,13 @@ class Engine {
   void init_gpu();
   void free_gpu();
   void join_tree();
-  bool try_connect(const sockaddr_in& addr, int& out_fd);
-  void handshake_as_child(int fd);
+  bool try_connect(const sockaddr_in& addr, int& out_fd,
+                   const sockaddr_in* bind_local = nullptr);
+  void handshake_as_child(int fd, bool rejoin = false);
+  void reconnect_loop();
+  void drop_children();
+  bool failover_master();
+  void zero_buf(float* p, int64_t n);
   void become_master();
   void bind_listen(const sockaddr_in& addr);
   void listen_loop();

@@ -1,0 +1,138 @@
+"""Fault tolerance: reconnection and master failover (the reference's
+acknowledged missing feature — any disconnect exits the process,
+/root/reference/src/sharedtensor.c:62,80,99 and README.md:33)."""
+import multiprocessing as mp
+import os
+import socket
+import time
+
+import pytest
+import torch
+
+import sharedtensor_amd as st
+
+
+def free_port():
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    p = s.getsockname()[1]
+    s.close()
+    return p
+
+
+def wait_until(fn, timeout=30.0, interval=0.05):
+    t0 = time.time()
+    while time.time() - t0 < timeout:
+        if fn():
+            return True
+        time.sleep(interval)
+    return False
+
+
+def _mortal_master(port, die_ev, ready_q):
+    h = st.create_or_fetch("127.0.0.1", port, torch.full((128,), 7.0))
+    ready_q.put(h.is_master)
+    die_ev.wait(60)
+    os._exit(1)  # simulate a crash: no CLOSE, no cleanup
+
+
+def test_child_survives_master_crash_and_takes_over():
+    port = free_port()
+    ctx = mp.get_context("spawn")
+    die_ev = ctx.Event()
+    ready_q = ctx.Queue()
+    mproc = ctx.Process(target=_mortal_master, args=(port, die_ev, ready_q))
+    mproc.start()
+    assert ready_q.get(timeout=30) is True
+
+    child = st.create_or_fetch("127.0.0.1", port, torch.zeros(128),
+                               reconnect=True, join_timeout_s=30)
+    try:
+        out = torch.zeros(128)
+
+        def got_state():
+            child.copy_to_tensor(out)
+            return torch.allclose(out, torch.full((128,), 7.0), atol=1e-2)
+
+        assert wait_until(got_state, timeout=20)
+
+        # crash the master
+        die_ev.set()
+        mproc.join(timeout=15)
+
+        # the child must detect the death, fail to reconnect, and take over
+        # the rendezvous address (failover master)
+        assert wait_until(lambda: child.is_master, timeout=30), \
+            f"no failover: stats={child.stats()}"
+
+        # state survived the failover
+        child.copy_to_tensor(out)
+        assert torch.allclose(out, torch.full((128,), 7.0), atol=1e-2)
+
+        # and the new master accepts fresh joiners with the inherited state
+        child.add_from_tensor(torch.ones(128))
+        joiner = st.create_or_fetch("127.0.0.1", port, torch.zeros(128))
+        try:
+            out2 = torch.zeros(128)
+
+            def joined():
+                joiner.copy_to_tensor(out2)
+                return torch.allclose(out2, torch.full((128,), 8.0), atol=1e-2)
+
+            assert wait_until(joined, timeout=20), out2[:4]
+        finally:
+            joiner.close()
+    finally:
+        child.close()
+
+
+def _flaky_child(port, q):
+    try:
+        h = st.create_or_fetch("127.0.0.1", port, torch.zeros(64),
+                               reconnect=True, join_timeout_s=30)
+        out = torch.zeros(64)
+
+        def conv(v):
+            h.copy_to_tensor(out)
+            return torch.allclose(out, torch.full((64,), v), atol=1e-2)
+
+        ok1 = wait_until(lambda: conv(3.0), timeout=20)
+        q.put(("stage1", ok1, None))
+        # wait for the parent to bounce; we must re-sync afterwards
+        ok2 = wait_until(lambda: conv(9.0), timeout=40)
+        q.put(("stage2", ok2, str(out[:4]) + " / " + h.stats()["last_error"]))
+        time.sleep(1)
+        h.close()
+    except Exception as e:  # pragma: no cover
+        q.put(("error", False, repr(e)))
+
+
+def test_child_rejoins_restarted_master():
+    """Master closes (clean) and restarts; a reconnect-enabled child rejoins
+    it and converges to the new state, carrying its own unsent residual."""
+    port = free_port()
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    m1 = st.create_or_fetch("127.0.0.1", port, torch.full((64,), 3.0))
+    child = ctx.Process(target=_flaky_child, args=(port, q))
+    child.start()
+    try:
+        stage, ok, info = q.get(timeout=60)
+        assert stage == "stage1" and ok, info
+        # hard-stop master #1 WITHOUT a clean close: drop the listener and
+        # sockets (simulates a crash while keeping this test in-process)
+        m1._eng.close()
+        # rebind quickly as the restarted master with different state; the
+        # child's failover race may grab the port first, in which case it
+        # becomes master and m2 joins it instead — either way the tree heals.
+        m2 = st.create_or_fetch("127.0.0.1", port, torch.full((64,), 9.0),
+                                join_timeout_s=30)
+        if not m2.is_master:
+            # child won the failover race; push the new target state through
+            m2.add_from_tensor(torch.full((64,), 6.0))  # 3 + 6 = 9
+        stage, ok, info = q.get(timeout=60)
+        assert stage == "stage2" and ok, info
+        m2.close()
+    finally:
+        child.join(timeout=30)
+    assert child.exitcode == 0
