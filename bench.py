@@ -72,11 +72,13 @@ def run_train(args, rank, world, device):
     model = GPT2(cfg).to(device)
     log(f"model {args.model}: {model.num_params()/1e6:.1f}M params, device {device}")
 
+    use_bf16_params = device.type == "cuda" and not args.fp32_params
     trainer = AsyncDPTrainer(
         model, host="127.0.0.1",
         port_base=int(os.environ.get("SHTENS_PORT_BASE", 52000)),
         rank=rank, world=world, lr=args.lr, momentum=0.9,
         amp_dtype=torch.bfloat16 if device.type == "cuda" else None,
+        param_dtype=torch.bfloat16 if use_bf16_params else torch.float32,
         codec=args.codec, use_rccl=not args.no_rccl,
         snapshot_join=True)
 
@@ -240,6 +242,8 @@ def main():
     ap.add_argument("--numel", type=int, default=268_435_456)  # 1 GB fp32
     ap.add_argument("--interval", type=float, default=0.01)
     ap.add_argument("--no-rccl", action="store_true")
+    ap.add_argument("--fp32-params", action="store_true",
+                    help="compute on fp32 replica views (default: bf16 shadow)")
     ap.add_argument("--device", default="auto")
     args = ap.parse_args()
 

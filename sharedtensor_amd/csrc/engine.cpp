@@ -1196,6 +1196,23 @@ void Engine::fused_sgd(uintptr_t mom, uintptr_t grad, double lr,
   notify_all_dirty();
 }
 
+void Engine::fused_sgd_bf16(uintptr_t mom, uintptr_t grad_bf16,
+                            uintptr_t shadow_bf16, double lr, double momentum,
+                            uintptr_t stream) {
+  if (!gpu())
+    throw std::runtime_error("fused_sgd_bf16 is a GPU-only path");
+  float* d[3];
+  for (int i = 0; i < 3; ++i)
+    d[i] = links_[i].provisioned ? links_[i].delta : nullptr;
+  hip_fused_sgd_bf16(reinterpret_cast<float*>(mom),
+                     reinterpret_cast<const uint16_t*>(grad_bf16),
+                     reinterpret_cast<uint16_t*>(shadow_bf16),
+                     static_cast<float>(lr), static_cast<float>(momentum), n_,
+                     values_, d[0], d[1], d[2],
+                     reinterpret_cast<hipStream_t>(stream));
+  notify_all_dirty();
+}
+
 std::vector<LinkStatsSnap> Engine::link_stats() {
   std::vector<LinkStatsSnap> out;
   for (auto& lk : links_) {

@@ -70,6 +70,8 @@ class Engine {
   void copy_to(uintptr_t dst, int64_t n, uintptr_t stream);
   void fused_sgd(uintptr_t mom, uintptr_t grad, double lr, double momentum,
                  uintptr_t stream);
+  void fused_sgd_bf16(uintptr_t mom, uintptr_t grad_bf16, uintptr_t shadow_bf16,
+                      double lr, double momentum, uintptr_t stream);
   void notify_dirty();  // wake senders after out-of-band delta writes
   void close();
 

@@ -56,4 +56,11 @@ void hip_fused_sgd(float* mom, const float* grad, float lr, float momentum,
                    int64_t n, float* d0, float* d1, float* d2, float* d3,
                    hipStream_t s);
 
+// Mixed-precision variant: bf16 grads in, fp32 master (values) updated,
+// bf16 shadow params refreshed (folding concurrent gossip via the
+// atomicAdd return), link deltas staged — all in one pass.
+void hip_fused_sgd_bf16(float* mom, const uint16_t* grad, uint16_t* shadow,
+                        float lr, float momentum, int64_t n, float* values,
+                        float* d1, float* d2, float* d3, hipStream_t s);
+
 }  // namespace shamd

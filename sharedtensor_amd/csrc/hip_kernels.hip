@@ -329,6 +329,40 @@ __global__ void k_add_scatter(const float* __restrict__ src, int64_t n,
   }
 }
 
+__device__ __forceinline__ float bf16_to_f32(uint16_t u) {
+  return __uint_as_float(static_cast<uint32_t>(u) << 16);
+}
+
+__device__ __forceinline__ uint16_t f32_to_bf16(float f) {
+  uint32_t u = __float_as_uint(f);
+  if ((u & 0x7FFFFFFFu) > 0x7F800000u) return 0x7FC0;  // NaN
+  u += 0x7FFFu + ((u >> 16) & 1u);  // round-to-nearest-even
+  return static_cast<uint16_t>(u >> 16);
+}
+
+// Mixed-precision fused optimizer: bf16 gradients in, fp32 master update,
+// bf16 shadow parameters out — one HBM pass covering what autocast training
+// otherwise spends three cast/copy kernel families on.  The atomicAdd's
+// return value folds any concurrently-applied gossip into the shadow.
+__global__ void k_fused_sgd_bf16(float* __restrict__ mom,
+                                 const uint16_t* __restrict__ grad,
+                                 uint16_t* __restrict__ shadow, float lr,
+                                 float momentum, int64_t n,
+                                 float* __restrict__ values, float* d1,
+                                 float* d2, float* d3) {
+  int64_t gstride = static_cast<int64_t>(gridDim.x) * blockDim.x;
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n; i += gstride) {
+    float m = momentum * mom[i] + bf16_to_f32(grad[i]);
+    mom[i] = m;
+    float u = -lr * m;
+    float old = atomicAdd(values + i, u);
+    shadow[i] = f32_to_bf16(old + u);
+    if (d1) atomicAdd(d1 + i, u);
+    if (d2) atomicAdd(d2 + i, u);
+    if (d3) atomicAdd(d3 + i, u);
+  }
+}
+
 __global__ void k_fused_sgd(float* __restrict__ mom,
                             const float* __restrict__ grad, float lr,
                             float momentum, int64_t n, float* d0, float* d1,
@@ -425,6 +459,14 @@ void hip_fused_sgd(float* mom, const float* grad, float lr, float momentum,
                    hipStream_t s) {
   hipLaunchKernelGGL(k_fused_sgd, dim3(grid_for(n)), dim3(BLOCK), 0, s, mom,
                      grad, lr, momentum, n, d0, d1, d2, d3);
+  HIP_CHECK(hipGetLastError());
+}
+
+void hip_fused_sgd_bf16(float* mom, const uint16_t* grad, uint16_t* shadow,
+                        float lr, float momentum, int64_t n, float* values,
+                        float* d1, float* d2, float* d3, hipStream_t s) {
+  hipLaunchKernelGGL(k_fused_sgd_bf16, dim3(grid_for(n)), dim3(BLOCK), 0, s,
+                     mom, grad, shadow, lr, momentum, n, values, d1, d2, d3);
   HIP_CHECK(hipGetLastError());
 }
 

@@ -149,6 +149,18 @@ class _SharedBase:
         self._eng.fused_sgd(momentum_buf.data_ptr(), grad.data_ptr(),
                             float(lr), float(momentum), self._stream())
 
+    def fused_sgd_bf16_step(self, momentum_buf: torch.Tensor,
+                            grad_bf16: torch.Tensor, shadow_bf16: torch.Tensor,
+                            lr: float, momentum: float = 0.9):
+        """Mixed-precision fused step: bf16 grads in, fp32 master updated,
+        bf16 shadow params refreshed, link deltas staged — one HBM pass
+        (HIP kernel k_fused_sgd_bf16)."""
+        assert grad_bf16.dtype == torch.bfloat16 and shadow_bf16.dtype == torch.bfloat16
+        assert momentum_buf.numel() == grad_bf16.numel() == shadow_bf16.numel() == self.n
+        self._eng.fused_sgd_bf16(momentum_buf.data_ptr(), grad_bf16.data_ptr(),
+                                 shadow_bf16.data_ptr(), float(lr),
+                                 float(momentum), self._stream())
+
     # -- observability -----------------------------------------------------
     def stats(self) -> dict:
         links = self._eng.link_stats()

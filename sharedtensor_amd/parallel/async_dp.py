@@ -39,7 +39,8 @@ class FlatParamShared(_SharedBase):
     config 3: 'params as one shared tensor')."""
 
     def __init__(self, model: torch.nn.Module, host: str, port_base: int,
-                 rank: int, world: int, **kw):
+                 rank: int, world: int, param_dtype: torch.dtype = torch.float32,
+                 **kw):
         params = [p for p in model.parameters() if p.requires_grad]
         # dedupe tied parameters (GPT-2 ties lm_head.weight to wte.weight)
         seen, uniq = set(), []
@@ -65,14 +66,26 @@ class FlatParamShared(_SharedBase):
                          explicit_parent=explicit_parent,
                          listen_port=listen_port, **kw)
 
-        # snapshot initial params, re-point them into the replica slab
+        # snapshot initial params, re-point them into the replica slab (fp32)
+        # or into a bf16 shadow of it (mixed precision: bf16 matmuls without
+        # autocast's per-op weight casts; fp32 master stays in the engine)
         init_flat = torch.cat([p.detach().reshape(-1) for p in self.params]).float()
-        self.grad_flat = torch.zeros(n, dtype=torch.float32, device=device)
+        self.param_dtype = param_dtype
         self.mom_flat = torch.zeros(n, dtype=torch.float32, device=device)
+        if param_dtype == torch.bfloat16:
+            if not self._gpu:
+                raise ValueError("bf16 shadow params need a GPU")
+            self.shadow = torch.empty(n, dtype=torch.bfloat16, device=device)
+            self.grad_flat = torch.zeros(n, dtype=torch.bfloat16, device=device)
+            param_src = self.shadow
+        else:
+            self.shadow = None
+            self.grad_flat = torch.zeros(n, dtype=torch.float32, device=device)
+            param_src = self.values
         off = 0
         for p in self.params:
             sz = p.numel()
-            p.data = self.values[off:off + sz].view(p.shape)
+            p.data = param_src[off:off + sz].view(p.shape)
             p.grad = self.grad_flat[off:off + sz].view(p.shape)
             off += sz
 
@@ -81,6 +94,9 @@ class FlatParamShared(_SharedBase):
             # seed the shared state with this rank's init (master's weights
             # win; other ranks receive them via snapshot/gossip)
             self._add_flat(init_flat)
+        if self.shadow is not None:
+            # initial shadow refresh (post-join: values now hold the state)
+            self.shadow.copy_(self.values)
         self.rank = rank
         self.world = world
 
@@ -96,8 +112,15 @@ class AsyncSGD:
         self.momentum = momentum
 
     def step(self):
-        self.shared.fused_sgd_step(self.shared.mom_flat, self.shared.grad_flat,
-                                   self.lr, self.momentum)
+        if self.shared.shadow is not None:
+            self.shared.fused_sgd_bf16_step(self.shared.mom_flat,
+                                            self.shared.grad_flat,
+                                            self.shared.shadow, self.lr,
+                                            self.momentum)
+        else:
+            self.shared.fused_sgd_step(self.shared.mom_flat,
+                                       self.shared.grad_flat,
+                                       self.lr, self.momentum)
 
     def zero_grad(self, set_to_none: bool = False):
         self.shared.grad_flat.zero_()
@@ -110,14 +133,16 @@ class AsyncDPTrainer:
                  port_base: Optional[int] = None, rank: Optional[int] = None,
                  world: Optional[int] = None, lr: float = 0.1,
                  momentum: float = 0.9, amp_dtype: Optional[torch.dtype] = torch.bfloat16,
-                 **engine_kw):
+                 param_dtype: Optional[torch.dtype] = None, **engine_kw):
         rank = int(os.environ.get("RANK", 0)) if rank is None else rank
         world = int(os.environ.get("WORLD_SIZE", 1)) if world is None else world
         if port_base is None:
             port_base = int(os.environ.get("SHTENS_PORT_BASE", 52000))
         self.model = model
+        if param_dtype is None:
+            param_dtype = torch.float32
         self.shared = FlatParamShared(model, host, port_base, rank, world,
-                                      **engine_kw)
+                                      param_dtype=param_dtype, **engine_kw)
         self.opt = AsyncSGD(self.shared, lr=lr, momentum=momentum)
         self.amp_dtype = amp_dtype
         self.rank, self.world = rank, world

@@ -110,6 +110,38 @@ def test_add_scatter_and_fused_sgd():
     torch.testing.assert_close(delta, u, rtol=1e-5, atol=1e-6)
 
 
+def test_fused_sgd_bf16():
+    n = 1 << 18
+    torch.manual_seed(11)
+    mom = torch.randn(n, device="cuda")
+    grad = torch.randn(n, device="cuda", dtype=torch.bfloat16)
+    vals = torch.randn(n, device="cuda")
+    delta = torch.zeros(n, device="cuda")
+    shadow = torch.zeros(n, device="cuda", dtype=torch.bfloat16)
+    mom_ref, vals_ref = mom.clone(), vals.clone()
+    lr, mu = 0.1, 0.9
+    # exercise via DevCodec-less direct binding through an engine-free call:
+    from sharedtensor_amd import _core
+    # use the plain binding via a tiny Engine is heavyweight; the kernel is
+    # covered through bench/smoke; here check semantics with the raw launcher
+    # exposed on Engine — fall back to building a master-only shared tensor.
+    import sharedtensor_amd as st
+    import socket
+    s = socket.socket(); s.bind(("127.0.0.1", 0)); port = s.getsockname()[1]; s.close()
+    h = st.create_or_fetch("127.0.0.1", port, vals_ref.clone())
+    h.fused_sgd_bf16_step(mom, grad, shadow, lr, mu)
+    out = torch.zeros(n, device="cuda")
+    h.copy_to_tensor(out)
+    torch.cuda.synchronize()
+    m_new = mu * mom_ref + grad.float()
+    u = -lr * m_new
+    torch.testing.assert_close(mom, m_new, rtol=1e-6, atol=1e-6)
+    torch.testing.assert_close(out, vals_ref + u, rtol=1e-5, atol=1e-6)
+    torch.testing.assert_close(shadow.float(), (vals_ref + u).to(torch.bfloat16).float(),
+                               rtol=0, atol=0)
+    h.close()
+
+
 def test_quantize_keepalive_zero_scale():
     n = 4096
     d = torch.zeros(n, device="cuda")
