@@ -80,6 +80,7 @@ def run_train(args, rank, world, device):
         amp_dtype=torch.bfloat16 if device.type == "cuda" else None,
         param_dtype=torch.bfloat16 if use_bf16_params else torch.float32,
         codec=args.codec, use_rccl=not args.no_rccl,
+        lagged_scale=args.lagged_scale and device.type == "cuda",
         snapshot_join=True)
 
     B, T = args.batch, cfg.block_size
@@ -171,6 +172,7 @@ def run_paramsync(args, rank, world, device):
     nchild = len(tree_children(rank, world))
     sh = _SharedBase(
         "127.0.0.1", port_base, [n], device=device, codec=args.codec,
+        lagged_scale=args.lagged_scale and device.type == "cuda",
         use_rccl=not args.no_rccl, expected_children=nchild if world > 1 else 0,
         provision_up=rank > 0,
         explicit_parent=f"127.0.0.1:{port_base + tree_parent(rank)}" if rank else "",
@@ -244,6 +246,8 @@ def main():
     ap.add_argument("--no-rccl", action="store_true")
     ap.add_argument("--fp32-params", action="store_true",
                     help="compute on fp32 replica views (default: bf16 shadow)")
+    ap.add_argument("--lagged-scale", action="store_true",
+                    help="fold scale stats into the quantize kernel")
     ap.add_argument("--device", default="auto")
     args = ap.parse_args()
 

@@ -82,11 +82,17 @@ struct DevCodec {
                       reinterpret_cast<hipStream_t>(stream));
   }
   void quantize(uintptr_t delta, uintptr_t scales_dev, uintptr_t payload,
-                uintptr_t stream) {
+                uintptr_t stream, uintptr_t stats = 0) {
     hip_quantize(c, reinterpret_cast<float*>(delta), tb,
                  reinterpret_cast<const float*>(scales_dev),
                  reinterpret_cast<uint8_t*>(payload),
-                 reinterpret_cast<hipStream_t>(stream));
+                 reinterpret_cast<hipStream_t>(stream),
+                 reinterpret_cast<void*>(stats));
+  }
+  void finalize_scales(uintptr_t stats, uintptr_t scales_dev, uintptr_t stream) {
+    hip_finalize_scales(c, tb, reinterpret_cast<const void*>(stats),
+                        reinterpret_cast<float*>(scales_dev), 1,
+                        reinterpret_cast<hipStream_t>(stream));
   }
   void apply(uintptr_t payload, uintptr_t scales_dev,
              std::vector<uintptr_t> dsts, uintptr_t stream) {
@@ -143,7 +149,8 @@ PYBIND11_MODULE(_core, m) {
       .def_readwrite("explicit_parent", &Config::explicit_parent)
       .def_readwrite("listen_port", &Config::listen_port)
       .def_readwrite("join_timeout_s", &Config::join_timeout_s)
-      .def_readwrite("rms_sample_stride", &Config::rms_sample_stride);
+      .def_readwrite("rms_sample_stride", &Config::rms_sample_stride)
+      .def_readwrite("lagged_scale", &Config::lagged_scale);
 
   py::class_<Engine>(m, "Engine")
       .def(py::init<Config>())
@@ -187,7 +194,10 @@ PYBIND11_MODULE(_core, m) {
   py::class_<DevCodec>(m, "DevCodec")
       .def(py::init<int, std::vector<int64_t>, int>())
       .def("reduce_scales", &DevCodec::reduce_scales)
-      .def("quantize", &DevCodec::quantize)
+      .def("quantize", &DevCodec::quantize, py::arg("delta"),
+           py::arg("scales"), py::arg("payload"), py::arg("stream"),
+           py::arg("stats") = 0)
+      .def("finalize_scales", &DevCodec::finalize_scales)
       .def("apply", &DevCodec::apply);
   m.def("gpu_add_scatter", &py_gpu_add_scatter);
   m.def("gpu_fused_sgd", &py_gpu_fused_sgd);

@@ -62,6 +62,10 @@ struct Config {
   double join_timeout_s = 60.0;
   int rms_sample_stride = 1;  // >1: subsample the RMS reduction (scale is a
                               // heuristic; stride k cuts its HBM traffic k×)
+  bool lagged_scale = false;  // GPU: fold the scale statistic into the
+                              // quantize kernel (next round's scale from this
+                              // round's post-quantize residual) — removes the
+                              // reduce pass from the steady state
 };
 
 struct LinkStatsSnap {

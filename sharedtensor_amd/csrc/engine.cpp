@@ -876,12 +876,22 @@ void Engine::recv_snapshot(int fd) {
 
 // -------------------------------------------------------------- send side
 
-void Engine::compute_scales(Link& lk, float* scales_host) {
+void Engine::compute_scales(Link& lk, float* scales_host, bool lagged_valid) {
   if (gpu()) {
     HIP_TRY(hipSetDevice(cfg_.device));
-    hip_reduce_scales(cfg_.codec, lk.delta, dtb_, lk.reduce_buf,
-                      reinterpret_cast<float*>(lk.send_buf),
-                      cfg_.rms_sample_stride, lk.s_send);
+    float* scales_dev = reinterpret_cast<float*>(lk.send_buf);
+    if (cfg_.lagged_scale && lagged_valid) {
+      // steady state: the stats were accumulated by the previous round's
+      // quantize kernel — no reduce pass over the residual needed
+      hip_finalize_scales(cfg_.codec, dtb_, lk.reduce_buf, scales_dev,
+                          /*stride=*/1, lk.s_send);
+      HIP_TRY(hipMemsetAsync(lk.reduce_buf, 0, 8 * T_, lk.s_send));
+    } else {
+      hip_reduce_scales(cfg_.codec, lk.delta, dtb_, lk.reduce_buf, scales_dev,
+                        cfg_.rms_sample_stride, lk.s_send);
+      if (cfg_.lagged_scale)
+        HIP_TRY(hipMemsetAsync(lk.reduce_buf, 0, 8 * T_, lk.s_send));
+    }
     HIP_TRY(hipMemcpyAsync(scales_host, lk.send_buf, 4 * T_,
                            hipMemcpyDeviceToHost, lk.s_send));
     HIP_TRY(hipStreamSynchronize(lk.s_send));
@@ -917,9 +927,10 @@ void Engine::send_loop(Link& lk) {
   auto last_send = Clock::now();
   auto next_allowed = Clock::now();
   const double keepalive = cfg_.keepalive_s;
+  bool lagged_valid = false;
   while (!closing_ && lk.state.load() == L_ACTIVE) {
     try {
-      compute_scales(lk, scales.data());
+      compute_scales(lk, scales.data(), lagged_valid);
     } catch (const std::exception& e) {
       link_down(lk, std::string("scale reduction failed: ") + e.what(), false);
       break;
@@ -941,6 +952,7 @@ void Engine::send_loop(Link& lk) {
         }
         last_send = now;
       }
+      lagged_valid = false;  // residual went quiet; restats on wake
       std::unique_lock<std::mutex> l(lk.m);
       lk.cv.wait_for(l, std::chrono::duration<double>(keepalive), [&] {
         return lk.dirty || closing_ || lk.state.load() != L_ACTIVE;
@@ -960,7 +972,8 @@ void Engine::send_loop(Link& lk) {
       if (gpu()) {
         hip_quantize(cfg_.codec, lk.delta, dtb_,
                      reinterpret_cast<float*>(lk.send_buf), lk.send_buf + SA_,
-                     lk.s_send);
+                     lk.s_send, cfg_.lagged_scale ? lk.reduce_buf : nullptr);
+        lagged_valid = true;
       } else {
         for (int t = 0; t < T_; ++t)
           cpu_quantize(cfg_.codec, lk.delta + offs_[t], cfg_.sizes[t], scales[t],

@@ -141,7 +141,7 @@ class Engine {
   void apply_packet(Link& lk, const float* scales_host);
   void link_down(Link& lk, const std::string& why, bool remote);
   void set_error(const std::string& e);
-  void compute_scales(Link& lk, float* scales_host);
+  void compute_scales(Link& lk, float* scales_host, bool lagged_valid);
   void notify_all_dirty();
   void push_scale(bool sent, float s);
 };

@@ -36,8 +36,16 @@ void hip_reduce_scales(Codec c, const float* delta, const DevTable& tb,
 
 // delta is debited in place (atomic, lossless vs concurrent adds);
 // payload receives the packed bytes for the whole padded space.
+// stats_out (nullable, 8*T bytes): accumulate post-quantize residual
+// statistics for lagged-scale mode (zero it before the call).
 void hip_quantize(Codec c, float* delta, const DevTable& tb,
-                  const float* scales_dev, uint8_t* payload, hipStream_t s);
+                  const float* scales_dev, uint8_t* payload, hipStream_t s,
+                  void* stats_out = nullptr);
+
+// Finalize scales from an already-populated reduce/stats buffer (lagged
+// mode: the buffer was filled by the previous round's quantize).
+void hip_finalize_scales(Codec c, const DevTable& tb, const void* reduce_buf,
+                         float* scales_out, int sample_stride, hipStream_t s);
 
 // Decode payload and accumulate into up to 4 destinations (values plus the
 // other links' delta buffers, sharedtensor.c:106-127).

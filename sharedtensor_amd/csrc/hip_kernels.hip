@@ -203,81 +203,135 @@ __global__ void k_finalize_scales(Codec c, const void* reduce_buf,
 // atomicAdd of -sent so concurrent adds (training thread, gossip forwards)
 // landing between the read and the update are preserved — the GPU version of
 // the reference's benign-race contract made lossless.
+// stats_out (nullable): accumulate the POST-quantize residual's sum of
+// squares per tensor (lagged-scale mode: next round's scale comes from this
+// round's quantize, so the separate reduce pass disappears from the steady
+// state).  Same register-accumulate + wave-flush structure as k_reduce_*.
 __global__ void k_quant_1bit(float* __restrict__ delta, const int64_t* offs,
                              const int64_t* poffs, int T, int64_t pe,
                              const float* __restrict__ scales,
-                             uint64_t* __restrict__ words) {
+                             uint64_t* __restrict__ words, double* stats_out) {
   int64_t gstride = static_cast<int64_t>(gridDim.x) * blockDim.x;
+  double acc = 0.0;
+  int cur_t = -1;
   for (int64_t j = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; j < pe; j += gstride) {
     int t = (T == 1) ? 0 : find_tensor(poffs, T, j);
+    if (stats_out && t != cur_t) {
+      if (cur_t >= 0) wave_flush_sumsq(acc, cur_t, stats_out);
+      acc = 0.0;
+      cur_t = t;
+    }
     int64_t L = j - poffs[t];
     int64_t sz = offs[t + 1] - offs[t];
     float s = scales[t];
     bool neg = false;  // bit value: 1 means -scale was sent
-    if (L < sz && s != 0.0f) {
+    if (L < sz) {
       float* p = delta + offs[t] + L;
       float v = __hip_atomic_load(reinterpret_cast<float*>(p), __ATOMIC_RELAXED,
                                   __HIP_MEMORY_SCOPE_AGENT);
-      neg = !(v > 0.0f);
-      atomicAdd(p, neg ? s : -s);
+      if (s != 0.0f) {
+        neg = !(v > 0.0f);
+        atomicAdd(p, neg ? s : -s);
+      }
+      if (stats_out) {
+        // fp32 subtraction first: the statistic must see the same rounded
+        // residual that lands in memory
+        float rf = (s == 0.0f) ? v : (v - (neg ? -s : s));
+        acc += static_cast<double>(rf) * static_cast<double>(rf);
+      }
     }
     uint64_t mask = __ballot(neg);
     if ((threadIdx.x & 63) == 0) words[j >> 6] = mask;
   }
+  if (stats_out && cur_t >= 0) wave_flush_sumsq(acc, cur_t, stats_out);
 }
 
 __global__ void k_quant_fp8(float* __restrict__ delta, const int64_t* offs,
                             const int64_t* poffs, int T, int64_t pe,
                             const float* __restrict__ scales,
-                            uint8_t* __restrict__ payload) {
+                            uint8_t* __restrict__ payload, uint32_t* stats_out) {
   int64_t gstride = static_cast<int64_t>(gridDim.x) * blockDim.x;
+  float amax = 0.0f;
+  int cur_t = -1;
   for (int64_t j = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; j < pe; j += gstride) {
     int t = (T == 1) ? 0 : find_tensor(poffs, T, j);
+    if (stats_out && t != cur_t) {
+      if (cur_t >= 0) wave_flush_max(amax, cur_t, stats_out);
+      amax = 0.0f;
+      cur_t = t;
+    }
     int64_t L = j - poffs[t];
     int64_t sz = offs[t + 1] - offs[t];
     float s = scales[t];
     uint8_t q = 0;
-    if (L < sz && s != 0.0f) {
+    if (L < sz) {
       float* p = delta + offs[t] + L;
       float v = __hip_atomic_load(p, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-      q = d_f32_to_e4m3(v / s);
-      atomicAdd(p, -d_e4m3_to_f32(q) * s);
+      float sent = 0.0f;
+      if (s != 0.0f) {
+        q = d_f32_to_e4m3(v / s);
+        sent = d_e4m3_to_f32(q) * s;
+        atomicAdd(p, -sent);
+      }
+      if (stats_out) {
+        float r = fabsf(v - sent);
+        amax = r > amax ? r : amax;
+      }
     }
     payload[j] = q;
   }
+  if (stats_out && cur_t >= 0) wave_flush_max(amax, cur_t, stats_out);
 }
 
 __global__ void k_quant_int4(float* __restrict__ delta, const int64_t* offs,
                              const int64_t* poffs, int T, int64_t pe,
                              const float* __restrict__ scales,
-                             uint8_t* __restrict__ payload) {
+                             uint8_t* __restrict__ payload, uint32_t* stats_out) {
   // one thread per payload byte = 2 elements (both in the same tensor:
-  // padded regions are 64-aligned, hence even)
+  // padded regions are 64-aligned, hence even).  nb is only a multiple of
+  // 32, so the loop runs to a 64-multiple with clamped indices to keep the
+  // wave fully convergent for the stats __shfl flush.
   int64_t nb = pe / 2;
+  int64_t nbp = (nb + 63) & ~int64_t(63);
   int64_t gstride = static_cast<int64_t>(gridDim.x) * blockDim.x;
-  for (int64_t b = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; b < nb; b += gstride) {
-    int64_t j0 = b * 2;
+  float amax = 0.0f;
+  int cur_t = -1;
+  for (int64_t b = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; b < nbp; b += gstride) {
+    bool act = b < nb;
+    int64_t j0 = (act ? b : nb - 1) * 2;
     int t = (T == 1) ? 0 : find_tensor(poffs, T, j0);
+    if (stats_out && t != cur_t) {
+      if (cur_t >= 0) wave_flush_max(amax, cur_t, stats_out);
+      amax = 0.0f;
+      cur_t = t;
+    }
     int64_t base = j0 - poffs[t];
     int64_t sz = offs[t + 1] - offs[t];
     float s = scales[t];
     uint8_t byte = 0;
-    if (s != 0.0f) {
-      for (int k = 0; k < 2; ++k) {
-        int64_t L = base + k;
-        if (L < sz) {
-          float* p = delta + offs[t] + L;
-          float v = __hip_atomic_load(p, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+    for (int k = 0; k < 2; ++k) {
+      int64_t L = base + k;
+      if (act && L < sz) {
+        float* p = delta + offs[t] + L;
+        float v = __hip_atomic_load(p, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+        float sent = 0.0f;
+        if (s != 0.0f) {
           float r = nearbyintf(v / s);
           r = r > 7.f ? 7.f : (r < -7.f ? -7.f : r);
           int8_t q = static_cast<int8_t>(r);
           byte |= static_cast<uint8_t>(q & 0xF) << (k * 4);
-          atomicAdd(p, -static_cast<float>(q) * s);
+          sent = static_cast<float>(q) * s;
+          atomicAdd(p, -sent);
+        }
+        if (stats_out) {
+          float r2 = fabsf(v - sent);
+          amax = r2 > amax ? r2 : amax;
         }
       }
     }
-    payload[b] = byte;
+    if (act) payload[b] = byte;
   }
+  if (stats_out && cur_t >= 0) wave_flush_max(amax, cur_t, stats_out);
 }
 
 // ------------------------------------------------------------------ apply
@@ -405,22 +459,35 @@ void hip_reduce_scales(Codec c, const float* delta, const DevTable& tb,
   HIP_CHECK(hipGetLastError());
 }
 
+void hip_finalize_scales(Codec c, const DevTable& tb, const void* reduce_buf,
+                         float* scales_out, int stride, hipStream_t s) {
+  int gt = (tb.T + BLOCK - 1) / BLOCK;
+  hipLaunchKernelGGL(k_finalize_scales, dim3(gt), dim3(BLOCK), 0, s, c,
+                     reduce_buf, tb.offs, tb.T, stride, scales_out);
+  HIP_CHECK(hipGetLastError());
+}
+
 void hip_quantize(Codec c, float* delta, const DevTable& tb,
-                  const float* scales_dev, uint8_t* payload, hipStream_t s) {
+                  const float* scales_dev, uint8_t* payload, hipStream_t s,
+                  void* stats_out) {
   int g = grid_for(c == Codec::Int4 ? tb.pe / 2 : tb.pe);
+  if (stats_out && g > 2048) g = 2048;  // bound the flush-atomic count
   switch (c) {
     case Codec::OneBit:
       hipLaunchKernelGGL(k_quant_1bit, dim3(g), dim3(BLOCK), 0, s, delta,
                          tb.offs, tb.poffs, tb.T, tb.pe, scales_dev,
-                         reinterpret_cast<uint64_t*>(payload));
+                         reinterpret_cast<uint64_t*>(payload),
+                         reinterpret_cast<double*>(stats_out));
       break;
     case Codec::Fp8:
       hipLaunchKernelGGL(k_quant_fp8, dim3(g), dim3(BLOCK), 0, s, delta,
-                         tb.offs, tb.poffs, tb.T, tb.pe, scales_dev, payload);
+                         tb.offs, tb.poffs, tb.T, tb.pe, scales_dev, payload,
+                         reinterpret_cast<uint32_t*>(stats_out));
       break;
     case Codec::Int4:
       hipLaunchKernelGGL(k_quant_int4, dim3(g), dim3(BLOCK), 0, s, delta,
-                         tb.offs, tb.poffs, tb.T, tb.pe, scales_dev, payload);
+                         tb.offs, tb.poffs, tb.T, tb.pe, scales_dev, payload,
+                         reinterpret_cast<uint32_t*>(stats_out));
       break;
   }
   HIP_CHECK(hipGetLastError());

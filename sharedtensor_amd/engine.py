@@ -41,7 +41,7 @@ class _SharedBase:
                  bw_limit: float = 0.0, expected_children: int = 2,
                  provision_up: bool = True, explicit_parent: str = "",
                  listen_port: int = 0, join_timeout_s: float = 60.0,
-                 rms_sample_stride: int = 1):
+                 rms_sample_stride: int = 1, lagged_scale: bool = False):
         if codec not in CODECS:
             raise ValueError(f"codec must be one of {list(CODECS)}")
         self.device = torch.device(device)
@@ -62,6 +62,7 @@ class _SharedBase:
         cfg.listen_port = int(listen_port)
         cfg.join_timeout_s = join_timeout_s
         cfg.rms_sample_stride = rms_sample_stride
+        cfg.lagged_scale = bool(lagged_scale)
         self.codec = codec
         self.n = int(sum(sizes))
         self._cfg = cfg
