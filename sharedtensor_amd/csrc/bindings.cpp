@@ -170,6 +170,7 @@ PYBIND11_MODULE(_core, m) {
       .def("is_master", &Engine::is_master)
       .def("listen_port", &Engine::listen_port)
       .def("last_error", &Engine::last_error)
+      .def("reconnect_count", &Engine::reconnect_count)
       .def("recent_scales_sent", &Engine::recent_scales_sent)
       .def("recent_scales_recv", &Engine::recent_scales_recv)
       .def("link_stats", [](Engine& e) {

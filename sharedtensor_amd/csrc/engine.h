@@ -81,6 +81,7 @@ class Engine {
   std::vector<float> recent_scales_sent();
   std::vector<float> recent_scales_recv();
   std::string last_error();
+  uint64_t reconnect_count() const { return reconnects_.load(); }
 
   // exposed sizes (python uses these to size buffers; static helpers)
   static int64_t msg_bytes(const Config& cfg);    // SA + P

@@ -180,6 +180,7 @@ class _SharedBase:
             "staleness_p50": _percentile(recv_scales, 0.5),
             "staleness_p90": _percentile(recv_scales, 0.9),
             "sent_scale_p50": _percentile(sent_scales, 0.5),
+            "reconnects": self._eng.reconnect_count(),
             "last_error": self._eng.last_error(),
         }
 
