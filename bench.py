@@ -20,9 +20,19 @@ import os
 import sys
 import time
 
-import torch
+# the contract is ONE JSON line on stdout from rank 0; gloo/torch print
+# banners to fd 1, so reroute fd 1 -> stderr and keep the real stdout for
+# the final JSON write
+_REAL_STDOUT = os.dup(1)
+os.dup2(2, 1)
+
+import torch  # noqa: E402
 
 sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+
+def emit(result: dict):
+    os.write(_REAL_STDOUT, (json.dumps(result) + "\n").encode())
 
 
 def log(msg):
@@ -80,7 +90,7 @@ def run_train(args, rank, world, device):
         amp_dtype=torch.bfloat16 if device.type == "cuda" else None,
         param_dtype=torch.bfloat16 if use_bf16_params else torch.float32,
         codec=args.codec, use_rccl=not args.no_rccl,
-        lagged_scale=args.lagged_scale and device.type == "cuda",
+        lagged_scale=(not args.exact_scale) and device.type == "cuda",
         snapshot_join=True)
 
     B, T = args.batch, cfg.block_size
@@ -158,7 +168,7 @@ def run_train(args, rank, world, device):
         dist.barrier()
         dist.destroy_process_group()
     if rank == 0:
-        print(json.dumps(result), flush=True)
+        emit(result)
 
 
 def run_paramsync(args, rank, world, device):
@@ -172,7 +182,7 @@ def run_paramsync(args, rank, world, device):
     nchild = len(tree_children(rank, world))
     sh = _SharedBase(
         "127.0.0.1", port_base, [n], device=device, codec=args.codec,
-        lagged_scale=args.lagged_scale and device.type == "cuda",
+        lagged_scale=(not args.exact_scale) and device.type == "cuda",
         use_rccl=not args.no_rccl, expected_children=nchild if world > 1 else 0,
         provision_up=rank > 0,
         explicit_parent=f"127.0.0.1:{port_base + tree_parent(rank)}" if rank else "",
@@ -227,7 +237,7 @@ def run_paramsync(args, rank, world, device):
         dist.barrier()
         dist.destroy_process_group()
     if rank == 0:
-        print(json.dumps(result), flush=True)
+        emit(result)
 
 
 def main():
@@ -235,7 +245,7 @@ def main():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=20)
     ap.add_argument("--warmup", type=int, default=5)
-    ap.add_argument("--batch", type=int, default=8)
+    ap.add_argument("--batch", type=int, default=32)
     ap.add_argument("--seq", type=int, default=1024)
     ap.add_argument("--lr", type=float, default=0.01)
     ap.add_argument("--codec", choices=["1bit", "fp8", "int4"], default="1bit")
@@ -246,8 +256,8 @@ def main():
     ap.add_argument("--no-rccl", action="store_true")
     ap.add_argument("--fp32-params", action="store_true",
                     help="compute on fp32 replica views (default: bf16 shadow)")
-    ap.add_argument("--lagged-scale", action="store_true",
-                    help="fold scale stats into the quantize kernel")
+    ap.add_argument("--exact-scale", action="store_true",
+                    help="per-round exact scale reduce (default: lagged, fused into quantize)")
     ap.add_argument("--device", default="auto")
     args = ap.parse_args()
 
