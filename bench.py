@@ -171,6 +171,82 @@ def run_train(args, rank, world, device):
         emit(result)
 
 
+def run_table(args, rank, world, device):
+    """Table-of-tensors sync on a Llama model's parameters (BASELINE
+    config 4: per-tensor scales)."""
+    from sharedtensor_amd.engine import SharedTable
+    from sharedtensor_amd.models.llama import Llama, LlamaConfig
+    from sharedtensor_amd.parallel.async_dp import tree_children, tree_parent
+
+    lcfg = {"tiny": LlamaConfig.tiny, "llama1b": LlamaConfig.llama_1b,
+            "llama8b": LlamaConfig.llama3_8b}.get(args.model, LlamaConfig.llama_1b)()
+    torch.manual_seed(7)
+    model = Llama(lcfg).to(device)
+    log(f"llama table: {model.num_params()/1e6:.0f}M params, "
+        f"{len(list(model.parameters()))} tensors")
+    port_base = int(os.environ.get("SHTENS_PORT_BASE", 52000))
+    nchild = len(tree_children(rank, world))
+    sh = SharedTable(
+        "127.0.0.1", port_base, {n: p.data for n, p in model.named_parameters()},
+        codec=args.codec,
+        lagged_scale=(not args.exact_scale) and device.type == "cuda",
+        use_rccl=not args.no_rccl, expected_children=nchild if world > 1 else 0,
+        provision_up=rank > 0,
+        explicit_parent=f"127.0.0.1:{port_base + tree_parent(rank)}" if rank else "",
+        listen_port=port_base + rank if world > 1 else 0)
+    dist = dist_setup(world)
+    n = sh.n
+    delta = torch.randn(n, dtype=torch.float32, device=device) * 0.001
+
+    def one_step():
+        sh.add_from_tensors(delta)
+        if device.type == "cuda":
+            torch.cuda.synchronize()
+        time.sleep(args.interval)
+
+    for _ in range(args.warmup):
+        one_step()
+    barrier(dist)
+    s0 = sh.stats()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        one_step()
+    time.sleep(1.0)
+    t1 = time.perf_counter()
+    barrier(dist)
+    s1 = sh.stats()
+    dt = max_over_ranks(dist, t1 - t0)
+    wire = sum_over_ranks(dist, float(s1["bytes_sent"] - s0["bytes_sent"] +
+                                      s1["bytes_recv"] - s0["bytes_recv"]))
+    rounds = sum_over_ranks(dist, float(s1["rounds_sent"] - s0["rounds_sent"]))
+    result = {
+        "metric": "table-of-tensors param-sync GB/s (llama)",
+        "value": round(rounds * n * 4 / dt / 1e9, 3),
+        "unit": "GB/s logical",
+        "n_gpus": world,
+        "steps": args.steps,
+        "warmup": args.warmup,
+        "ms_per_step": round(dt / args.steps * 1e3, 3),
+        "higher_is_better": True,
+        "scaling": "weak",
+        "vs_baseline": None,
+        "dtype": "fp32",
+        "data": "synthetic",
+        "config": {
+            "model": args.model, "numel": n,
+            "n_tensors": len(sh.names), "codec": args.codec,
+            "wire_gbps": round(wire / dt / 1e9, 3),
+            "staleness_p50": s1["staleness_p50"],
+        },
+    }
+    sh.close()
+    if dist is not None:
+        dist.barrier()
+        dist.destroy_process_group()
+    if rank == 0:
+        emit(result)
+
+
 def run_paramsync(args, rank, world, device):
     """Secondary mode: raw param-sync bandwidth on a flat tensor
     (BASELINE configs 2 and 5)."""
@@ -249,8 +325,8 @@ def main():
     ap.add_argument("--seq", type=int, default=1024)
     ap.add_argument("--lr", type=float, default=0.01)
     ap.add_argument("--codec", choices=["1bit", "fp8", "int4"], default="1bit")
-    ap.add_argument("--model", choices=["small", "tiny"], default="small")
-    ap.add_argument("--mode", choices=["train", "paramsync"], default="train")
+    ap.add_argument("--model", choices=["small", "tiny", "llama1b", "llama8b"], default="small")
+    ap.add_argument("--mode", choices=["train", "paramsync", "table"], default="train")
     ap.add_argument("--numel", type=int, default=268_435_456)  # 1 GB fp32
     ap.add_argument("--interval", type=float, default=0.01)
     ap.add_argument("--no-rccl", action="store_true")
@@ -278,6 +354,8 @@ def main():
 
     if args.mode == "train":
         run_train(args, rank, world, device)
+    elif args.mode == "table":
+        run_table(args, rank, world, device)
     else:
         run_paramsync(args, rank, world, device)
 

@@ -9,6 +9,7 @@
 
 #include "codec_cpu.h"
 #include "engine.h"
+#include "rccl_transport.h"
 
 namespace py = pybind11;
 using namespace shamd;
@@ -202,6 +203,8 @@ PYBIND11_MODULE(_core, m) {
       .def("apply", &DevCodec::apply);
   m.def("gpu_add_scatter", &py_gpu_add_scatter);
   m.def("gpu_fused_sgd", &py_gpu_fused_sgd);
+  m.def("rccl_self_test", &rccl_self_test,
+        py::call_guard<py::gil_scoped_release>());
 
   m.def("msg_bytes", &Engine::msg_bytes);
   m.def("scales_area", &Engine::scales_area);

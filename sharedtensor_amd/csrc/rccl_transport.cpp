@@ -84,6 +84,27 @@ RcclLink* rccl_link_create(int device, const uint8_t ids[2 * RCCL_ID_BYTES],
   return l;
 }
 
+// Our communicators are non-blocking (blocking=0), so ncclSend/ncclRecv may
+// return ncclInProgress BEFORE the operation is enqueued on the stream —
+// polling the stream alone would report "done" on an empty stream and let
+// the caller reuse the buffer.  Order matters: (1) poll the comm's async
+// state until the op is enqueued, (2) then poll the stream until it drains.
+static bool wait_enqueued(RcclLink* l, ncclComm_t c,
+                          const std::atomic<bool>& abort) {
+  for (;;) {
+    ncclResult_t st;
+    if (ncclCommGetAsyncError(c, &st) != ncclSuccess) return false;
+    if (st == ncclSuccess) return true;
+    if (st != ncclInProgress) return false;
+    if (abort.load() || l->aborted.load()) {
+      ncclCommAbort(c);
+      l->aborted.store(true);
+      return false;
+    }
+    std::this_thread::sleep_for(std::chrono::microseconds(20));
+  }
+}
+
 static bool wait_stream(RcclLink* l, ncclComm_t c, hipStream_t s,
                         const std::atomic<bool>& abort) {
   for (;;) {
@@ -106,16 +127,18 @@ static bool wait_stream(RcclLink* l, ncclComm_t c, hipStream_t s,
 bool rccl_send(RcclLink* l, const void* buf, size_t bytes, hipStream_t stream,
                const std::atomic<bool>& abort) {
   if (l->aborted.load()) return false;
-  if (ncclSend(buf, bytes, ncclChar, l->peer, l->out, stream) != ncclSuccess)
-    return false;
+  ncclResult_t r = ncclSend(buf, bytes, ncclChar, l->peer, l->out, stream);
+  if (r != ncclSuccess && r != ncclInProgress) return false;
+  if (!wait_enqueued(l, l->out, abort)) return false;
   return wait_stream(l, l->out, stream, abort);
 }
 
 bool rccl_recv(RcclLink* l, void* buf, size_t bytes, hipStream_t stream,
                const std::atomic<bool>& abort) {
   if (l->aborted.load()) return false;
-  if (ncclRecv(buf, bytes, ncclChar, l->peer, l->in, stream) != ncclSuccess)
-    return false;
+  ncclResult_t r = ncclRecv(buf, bytes, ncclChar, l->peer, l->in, stream);
+  if (r != ncclSuccess && r != ncclInProgress) return false;
+  if (!wait_enqueued(l, l->in, abort)) return false;
   return wait_stream(l, l->in, stream, abort);
 }
 
@@ -123,6 +146,38 @@ void rccl_abort(RcclLink* l) {
   if (!l || l->aborted.exchange(true)) return;
   if (l->out) ncclCommAbort(l->out);
   if (l->in) ncclCommAbort(l->in);
+}
+
+void rccl_self_test(int device) {
+  if (hipSetDevice(device) != hipSuccess)
+    throw std::runtime_error("hipSetDevice failed");
+  ncclUniqueId id;
+  NCCL_TRY(ncclGetUniqueId(&id));
+  ncclComm_t c = nullptr;
+  ncclConfig_t cfg = NCCL_CONFIG_INITIALIZER;
+  cfg.blocking = 0;
+  NCCL_TRY(ncclCommInitRankConfig(&c, 1, id, 0, &cfg));
+  wait_comm(c, 30.0, "self-test init");
+  float* buf = nullptr;
+  if (hipMalloc(&buf, 1024 * 4) != hipSuccess)
+    throw std::runtime_error("hipMalloc failed");
+  hipMemset(buf, 0, 1024 * 4);
+  hipStream_t s;
+  hipStreamCreate(&s);
+  ncclResult_t r = ncclAllReduce(buf, buf, 1024, ncclFloat, ncclSum, c, s);
+  if (r != ncclSuccess && r != ncclInProgress) {
+    hipFree(buf);
+    throw std::runtime_error(std::string("self-test allreduce: ") +
+                             ncclGetErrorString(r));
+  }
+  wait_comm(c, 30.0, "self-test allreduce enqueue");
+  if (hipStreamSynchronize(s) != hipSuccess) {
+    hipFree(buf);
+    throw std::runtime_error("self-test stream sync failed");
+  }
+  hipStreamDestroy(s);
+  hipFree(buf);
+  ncclCommDestroy(c);
 }
 
 void rccl_destroy(RcclLink* l) {

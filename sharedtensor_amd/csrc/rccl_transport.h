@@ -41,4 +41,9 @@ bool rccl_recv(RcclLink* l, void* buf, size_t bytes, hipStream_t stream,
 void rccl_abort(RcclLink* l);    // break in-flight ops (idempotent)
 void rccl_destroy(RcclLink* l);  // abort + free
 
+// Single-GPU sanity check: size-1 non-blocking communicator + self
+// all-reduce.  Validates librccl + the non-blocking poll path without a
+// second GPU; throws on failure.
+void rccl_self_test(int device);
+
 }  // namespace shamd
