@@ -68,8 +68,11 @@ static void set_sockopts(int fd) {
   setsockopt(fd, IPPROTO_TCP, TCP_NODELAY, &yes, sizeof(yes));
   // on the up-connection this is what later allows binding our listener to
   // the same local (ip, port) — the reference's self-addressing trick
-  // (sharedtensor.c:264,292-316)
+  // (sharedtensor.c:264,292-316).  SO_REUSEPORT additionally lets a REJOIN
+  // bind its outgoing socket to the address our listener already occupies
+  // (REUSEADDR alone does not permit bind beside a LISTEN socket on Linux).
   setsockopt(fd, SOL_SOCKET, SO_REUSEADDR, &yes, sizeof(yes));
+  setsockopt(fd, SOL_SOCKET, SO_REUSEPORT, &yes, sizeof(yes));
   int buf = 8 << 20;
   setsockopt(fd, SOL_SOCKET, SO_SNDBUF, &buf, sizeof(buf));
   setsockopt(fd, SOL_SOCKET, SO_RCVBUF, &buf, sizeof(buf));
@@ -343,7 +346,7 @@ void Engine::join_tree() {
 }
 
 void Engine::become_master() {
-  bind_listen(root_addr_);
+  bind_listen(root_addr_, /*shared=*/false);
   is_master_ = true;
   std::fprintf(stderr,
                "[sharedtensor_amd] master tensor at %s (n=%lld, %d tensor%s)\n",
@@ -461,7 +464,7 @@ bool Engine::failover_master() {
   if (listen_thread_.joinable()) listen_thread_.join();
   if (listen_fd_ >= 0) ::close(listen_fd_), listen_fd_ = -1;
   try {
-    bind_listen(root_addr_);
+    bind_listen(root_addr_, /*shared=*/false);
     is_master_ = true;
   } catch (const std::exception&) {
     try {
@@ -601,11 +604,15 @@ void Engine::reconnect_loop() try {
   reconnecting_.store(false);
 }
 
-void Engine::bind_listen(const sockaddr_in& addr) {
+void Engine::bind_listen(const sockaddr_in& addr, bool shared) {
   listen_fd_ = ::socket(AF_INET, SOCK_STREAM, 0);
   if (listen_fd_ < 0) throw std::runtime_error("socket() failed");
   int yes = 1;
   setsockopt(listen_fd_, SOL_SOCKET, SO_REUSEADDR, &yes, sizeof(yes));
+  // child listeners are REUSEPORT so a rejoin can bind its outgoing socket
+  // beside them; the MASTER/rendezvous bind stays exclusive — bind() is the
+  // failover arbiter and REUSEPORT would allow split-brain
+  if (shared) setsockopt(listen_fd_, SOL_SOCKET, SO_REUSEPORT, &yes, sizeof(yes));
   if (::bind(listen_fd_, reinterpret_cast<const sockaddr*>(&addr), sizeof(addr)) < 0) {
     std::string err = std::strerror(errno);
     ::close(listen_fd_);

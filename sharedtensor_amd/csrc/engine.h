@@ -127,7 +127,7 @@ class Engine {
   bool failover_master();
   void zero_buf(float* p, int64_t n);
   void become_master();
-  void bind_listen(const sockaddr_in& addr);
+  void bind_listen(const sockaddr_in& addr, bool shared = true);
   void listen_loop();
   void accept_child(int fd, const Hello& h, const sockaddr_in& peer, int slot);
   void spawn_link_threads(Link& lk);
