@@ -20,6 +20,9 @@ import os
 import sys
 import time
 
+# multi-process GPU tensor sharing / RCCL need dmabuf IPC on this pool
+os.environ.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
+
 # the contract is ONE JSON line on stdout from rank 0; gloo/torch print
 # banners to fd 1, so reroute fd 1 -> stderr and keep the real stdout for
 # the final JSON write

@@ -234,6 +234,22 @@ class SharedTensor(_SharedBase):
         reads are the async-approximate contract, README.md:20-24)."""
         return self.values.view(self.shape)
 
+    # checkpointing: the reference's only "restore" is joining (the
+    # accumulated-delta bootstrap); we add explicit save/restore on top
+    def save(self, path: str):
+        torch.save({"values": self.values.detach().cpu(),
+                    "shape": self.shape, "codec": self.codec}, path)
+
+    @classmethod
+    def restore(cls, host: str, port: int, path: str,
+                device="cpu", **kw) -> "SharedTensor":
+        """Create/join a shared tensor seeded from a checkpoint.  If this
+        process becomes master, the checkpointed values seed the tree;
+        otherwise the live tree state wins (by reference join semantics)."""
+        ckpt = torch.load(path, map_location="cpu", weights_only=True)
+        seed = ckpt["values"].view(ckpt["shape"]).to(device)
+        return cls(host, port, seed, **kw)
+
 
 class SharedTable(_SharedBase):
     """Table-of-tensors sync with per-tensor scales (reference README.md:41).
