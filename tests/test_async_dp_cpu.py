@@ -85,3 +85,25 @@ def test_two_rank_async_dp_cpu():
             p.join(timeout=60)
     for p in procs:
         assert p.exitcode == 0
+
+
+def test_four_rank_async_dp_cpu():
+    """Depth-2 explicit tree (rank 0 <- 1,2; rank 1 <- 3): the same topology
+    wiring the driver's 8-GPU scaling bench uses, exercised on CPU."""
+    port_base = free_port_base()
+    assert port_base
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_worker, args=(r, 4, port_base, q))
+             for r in range(4)]
+    for p in procs:
+        p.start()
+    try:
+        results = [q.get(timeout=240) for _ in procs]
+        for status, msg in results:
+            assert status == "ok", msg
+    finally:
+        for p in procs:
+            p.join(timeout=60)
+    for p in procs:
+        assert p.exitcode == 0
