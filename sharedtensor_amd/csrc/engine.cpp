@@ -666,7 +666,14 @@ void Engine::listen_loop() {
     timeval tv0{0, 0};
     setsockopt(fd, SOL_SOCKET, SO_RCVTIMEO, &tv0, sizeof(tv0));
 
-    // pick a free child slot
+    if (reconnecting_.load()) {
+      // mid-rejoin our state is being reconciled; drop the joiner (it
+      // retries its walk) rather than snapshot half-rebuilt values
+      ::close(fd);
+      continue;
+    }
+    // pick a free child slot (serialized against reconnect demotion)
+    std::lock_guard<std::mutex> slot_guard(slots_m_);
     int slot = -1;
     for (int i : {LK_LEFT, LK_RIGHT}) {
       Link& lk = links_[i];
