@@ -37,6 +37,26 @@ class GPT2Config:
                    block_size=64)
 
 
+class FusedLayerNorm(nn.LayerNorm):
+    """LayerNorm that stays in bf16 under autocast.
+
+    torch's autocast policy runs LayerNorm in fp32, casting the whole
+    (B, T, C) activation up and back down every call — measured at ~9% of a
+    GPT-2-small step on MI355X (profiles/gpt2_train_step_kernels.txt).  The
+    native layer_norm kernel already accumulates statistics in fp32, so when
+    weights and input share a low-precision dtype we bypass the autocast
+    upcast entirely.
+    """
+
+    def forward(self, x):
+        if (x.is_cuda and x.dtype == torch.bfloat16
+                and self.weight.dtype == torch.bfloat16):
+            with torch.autocast("cuda", enabled=False):
+                return F.layer_norm(x, self.normalized_shape, self.weight,
+                                    self.bias, self.eps)
+        return super().forward(x)
+
+
 class CausalSelfAttention(nn.Module):
     def __init__(self, cfg: GPT2Config):
         super().__init__()
@@ -70,9 +90,9 @@ class MLP(nn.Module):
 class Block(nn.Module):
     def __init__(self, cfg: GPT2Config):
         super().__init__()
-        self.ln1 = nn.LayerNorm(cfg.n_embd)
+        self.ln1 = FusedLayerNorm(cfg.n_embd)
         self.attn = CausalSelfAttention(cfg)
-        self.ln2 = nn.LayerNorm(cfg.n_embd)
+        self.ln2 = FusedLayerNorm(cfg.n_embd)
         self.mlp = MLP(cfg)
 
     def forward(self, x):
@@ -88,7 +108,7 @@ class GPT2(nn.Module):
         self.wte = nn.Embedding(cfg.vocab_size, cfg.n_embd)
         self.wpe = nn.Embedding(cfg.block_size, cfg.n_embd)
         self.blocks = nn.ModuleList(Block(cfg) for _ in range(cfg.n_layer))
-        self.ln_f = nn.LayerNorm(cfg.n_embd)
+        self.ln_f = FusedLayerNorm(cfg.n_embd)
         self.lm_head = nn.Linear(cfg.n_embd, cfg.vocab_size, bias=False)
         self.lm_head.weight = self.wte.weight  # weight tying
         self.apply(self._init)
