@@ -253,13 +253,13 @@ def run_table(args, rank, world, device):
 def run_paramsync(args, rank, world, device):
     """Secondary mode: raw param-sync bandwidth on a flat tensor
     (BASELINE configs 2 and 5)."""
-    from sharedtensor_amd.engine import _SharedBase
+    from sharedtensor_amd.engine import SharedFlat
     from sharedtensor_amd.parallel.async_dp import tree_children, tree_parent
 
     n = args.numel
     port_base = int(os.environ.get("SHTENS_PORT_BASE", 52000))
     nchild = len(tree_children(rank, world))
-    sh = _SharedBase(
+    sh = SharedFlat(
         "127.0.0.1", port_base, [n], device=device, codec=args.codec,
         lagged_scale=(not args.exact_scale) and device.type == "cuda",
         use_rccl=not args.no_rccl, expected_children=nchild if world > 1 else 0,

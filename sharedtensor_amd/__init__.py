@@ -15,13 +15,14 @@ AMD Instinct MI355X (gfx950):
 Public API: SharedTensor / SharedTable / create_or_fetch (reference parity),
 models.* (GPT-2 flagship workload), parallel.* (async-DP trainer).
 """
-from .engine import (CODECS, SharedTable, SharedTensor, create_or_fetch,
-                     createOrFetch)
+from .engine import (CODECS, SharedFlat, SharedTable, SharedTensor,
+                     create_or_fetch, createOrFetch)
 
 __version__ = "0.1.0"
 
 __all__ = [
     "SharedTensor",
+    "SharedFlat",
     "SharedTable",
     "create_or_fetch",
     "createOrFetch",

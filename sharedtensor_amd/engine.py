@@ -321,6 +321,11 @@ class SharedTable(_SharedBase):
         return cls(host, port, params, **kw)
 
 
+# public name for the flat multi-tensor-capable base (bench/paramsync use it
+# directly when no shape semantics are needed)
+SharedFlat = _SharedBase
+
+
 def create_or_fetch(host: str, port: int, tensor: torch.Tensor, **kw) -> SharedTensor:
     """Reference entrypoint parity (sharedtensor.createOrFetch,
     sharedtensor.c:347-391)."""
