@@ -33,11 +33,12 @@ def wait_until(fn, timeout=30.0, interval=0.05):
     return False
 
 
-def _gpu_child(port, q, codec):
+def _gpu_child(port, q, codec, lagged=False):
     try:
         torch.cuda.set_device(0)
         seed = torch.zeros(1 << 20, device="cuda")
-        h = st.create_or_fetch("127.0.0.1", port, seed, codec=codec)
+        h = st.create_or_fetch("127.0.0.1", port, seed, codec=codec,
+                               lagged_scale=lagged)
         target = torch.full((1 << 20,), 3.0, device="cuda")
         out = torch.zeros_like(seed)
 
@@ -57,16 +58,17 @@ def _gpu_child(port, q, codec):
         q.put(("fail", repr(e)))
 
 
-@pytest.mark.parametrize("codec", ["1bit", "fp8"])
-def test_two_process_one_gpu_tcp(codec):
+@pytest.mark.parametrize("codec,lagged", [("1bit", False), ("fp8", False),
+                                          ("1bit", True), ("int4", True)])
+def test_two_process_one_gpu_tcp(codec, lagged):
     port = free_port()
     ctx = mp.get_context("spawn")
     torch.cuda.set_device(0)
     master = st.create_or_fetch("127.0.0.1", port,
                                 torch.full((1 << 20,), 3.0, device="cuda"),
-                                codec=codec)
+                                codec=codec, lagged_scale=lagged)
     q = ctx.Queue()
-    p = ctx.Process(target=_gpu_child, args=(port, q, codec))
+    p = ctx.Process(target=_gpu_child, args=(port, q, codec, lagged))
     p.start()
     try:
         status, msg = q.get(timeout=120)
