@@ -151,7 +151,8 @@ PYBIND11_MODULE(_core, m) {
       .def_readwrite("listen_port", &Config::listen_port)
       .def_readwrite("join_timeout_s", &Config::join_timeout_s)
       .def_readwrite("rms_sample_stride", &Config::rms_sample_stride)
-      .def_readwrite("lagged_scale", &Config::lagged_scale);
+      .def_readwrite("lagged_scale", &Config::lagged_scale)
+      .def_readwrite("use_graphs", &Config::use_graphs);
 
   py::class_<Engine>(m, "Engine")
       .def(py::init<Config>())

@@ -66,6 +66,9 @@ struct Config {
                               // quantize kernel (next round's scale from this
                               // round's post-quantize residual) — removes the
                               // reduce pass from the steady state
+  bool use_graphs = true;     // GPU: capture the per-round kernel+copy
+                              // sequences into hipGraphs (one replay instead
+                              // of 2-4 launches per round)
 };
 
 struct LinkStatsSnap {

@@ -30,6 +30,21 @@ using Clock = std::chrono::steady_clock;
                                std::to_string(__LINE__));                    \
   } while (0)
 
+// Capture a stream-op sequence into an executable hipGraph (one replay per
+// round instead of several launches).  ThreadLocal mode: other link threads'
+// streams keep running during capture.
+template <typename F>
+static hipGraphExec_t capture_seq(hipStream_t s, F&& body) {
+  hipGraph_t g = nullptr;
+  HIP_TRY(hipStreamBeginCapture(s, hipStreamCaptureModeThreadLocal));
+  body();
+  HIP_TRY(hipStreamEndCapture(s, &g));
+  hipGraphExec_t e = nullptr;
+  HIP_TRY(hipGraphInstantiate(&e, g, nullptr, nullptr, 0));
+  (void)hipGraphDestroy(g);
+  return e;
+}
+
 // ------------------------------------------------------------ I/O helpers
 // Loop partial reads/writes and retry EINTR (reference read_or_die/
 // write_or_die, sharedtensor.c:53-87) — but surface failure instead of
@@ -220,9 +235,17 @@ void Engine::init_gpu() {
   }
 }
 
+static void destroy_link_graphs(Link& lk) {
+  for (hipGraphExec_t* e : {&lk.g_scale_lagged, &lk.g_scale_exact, &lk.g_quant,
+                            &lk.g_apply}) {
+    if (*e) (void)hipGraphExecDestroy(*e), *e = nullptr;
+  }
+}
+
 void Engine::free_gpu() {
   if (!gpu()) return;
   for (auto& lk : links_) {
+    destroy_link_graphs(lk);
     if (lk.s_send) hipStreamDestroy(lk.s_send), lk.s_send = nullptr;
     if (lk.s_recv) hipStreamDestroy(lk.s_recv), lk.s_recv = nullptr;
     if (lk.reduce_buf) hipFree(lk.reduce_buf), lk.reduce_buf = nullptr;
@@ -451,6 +474,7 @@ void Engine::drop_children() {
     }
     lk.abort.store(false);
     lk.error.clear();
+    if (gpu()) destroy_link_graphs(lk);  // transport may change on reuse
     if (lk.provisioned) zero_buf(lk.delta, n_);
     lk.state.store(L_FREE);
   }
@@ -496,6 +520,7 @@ void Engine::reconnect_loop() try {
     up.rccl_link = nullptr;
     up.rccl = false;
   }
+  if (gpu()) destroy_link_graphs(up);
   up.abort.store(false);
   up.state.store(L_FREE);
   drop_children();
@@ -692,6 +717,7 @@ void Engine::listen_loop() {
           lk.rccl_link = nullptr;
           lk.rccl = false;
         }
+        if (gpu()) destroy_link_graphs(lk);
         lk.abort.store(false);
         lk.fd = -1;
         lk.error.clear();
@@ -894,21 +920,34 @@ void Engine::compute_scales(Link& lk, float* scales_host, bool lagged_valid) {
   if (gpu()) {
     HIP_TRY(hipSetDevice(cfg_.device));
     float* scales_dev = reinterpret_cast<float*>(lk.send_buf);
-    if (cfg_.lagged_scale && lagged_valid) {
-      // steady state: the stats were accumulated by the previous round's
-      // quantize kernel — no reduce pass over the residual needed
-      hip_finalize_scales(cfg_.codec, dtb_, lk.reduce_buf, scales_dev,
-                          /*stride=*/1, lk.s_send);
-      HIP_TRY(hipMemsetAsync(lk.reduce_buf, 0, 8 * T_, lk.s_send));
-    } else {
-      hip_reduce_scales(cfg_.codec, lk.delta, dtb_, lk.reduce_buf, scales_dev,
-                        cfg_.rms_sample_stride, lk.s_send);
-      if (cfg_.lagged_scale)
+    bool lagged = cfg_.lagged_scale && lagged_valid;
+    if (lk.scales_host.size() != static_cast<size_t>(T_))
+      lk.scales_host.resize(T_);
+    auto seq = [&] {
+      if (lagged) {
+        // steady state: the stats were accumulated by the previous round's
+        // quantize kernel — no reduce pass over the residual needed
+        hip_finalize_scales(cfg_.codec, dtb_, lk.reduce_buf, scales_dev,
+                            /*stride=*/1, lk.s_send);
         HIP_TRY(hipMemsetAsync(lk.reduce_buf, 0, 8 * T_, lk.s_send));
+      } else {
+        hip_reduce_scales(cfg_.codec, lk.delta, dtb_, lk.reduce_buf,
+                          scales_dev, cfg_.rms_sample_stride, lk.s_send);
+        if (cfg_.lagged_scale)
+          HIP_TRY(hipMemsetAsync(lk.reduce_buf, 0, 8 * T_, lk.s_send));
+      }
+      HIP_TRY(hipMemcpyAsync(lk.scales_host.data(), lk.send_buf, 4 * T_,
+                             hipMemcpyDeviceToHost, lk.s_send));
+    };
+    if (cfg_.use_graphs) {
+      hipGraphExec_t& e = lagged ? lk.g_scale_lagged : lk.g_scale_exact;
+      if (!e) e = capture_seq(lk.s_send, seq);
+      HIP_TRY(hipGraphLaunch(e, lk.s_send));
+    } else {
+      seq();
     }
-    HIP_TRY(hipMemcpyAsync(scales_host, lk.send_buf, 4 * T_,
-                           hipMemcpyDeviceToHost, lk.s_send));
     HIP_TRY(hipStreamSynchronize(lk.s_send));
+    std::memcpy(scales_host, lk.scales_host.data(), 4 * T_);
   } else {
     for (int t = 0; t < T_; ++t)
       scales_host[t] = cpu_compute_scale(cfg_.codec, lk.delta + offs_[t],
@@ -922,13 +961,10 @@ bool Engine::send_packet(Link& lk, const float* scales_host) {
   hdr.type = PKT_DATA;
   hdr.codec = static_cast<uint8_t>(cfg_.codec);
   hdr.ntensors = static_cast<uint32_t>(T_);
+  std::memcpy(lk.send_pin, &hdr, 8);
   if (gpu()) {
-    std::memcpy(lk.send_pin, &hdr, 8);
-    HIP_TRY(hipMemcpyAsync(lk.send_pin + 8, lk.send_buf, SA_ + P_,
-                           hipMemcpyDeviceToHost, lk.s_send));
+    // the payload D2H was part of the quantize phase (graph); just drain
     HIP_TRY(hipStreamSynchronize(lk.s_send));
-  } else {
-    std::memcpy(lk.send_pin, &hdr, 8);
   }
   std::lock_guard<std::mutex> g(lk.wm);
   if (!io_write(lk.fd, lk.send_pin, 8 + SA_ + P_)) return false;
@@ -984,9 +1020,20 @@ void Engine::send_loop(Link& lk) {
     }
     try {
       if (gpu()) {
-        hip_quantize(cfg_.codec, lk.delta, dtb_,
-                     reinterpret_cast<float*>(lk.send_buf), lk.send_buf + SA_,
-                     lk.s_send, cfg_.lagged_scale ? lk.reduce_buf : nullptr);
+        auto qseq = [&] {
+          hip_quantize(cfg_.codec, lk.delta, dtb_,
+                       reinterpret_cast<float*>(lk.send_buf), lk.send_buf + SA_,
+                       lk.s_send, cfg_.lagged_scale ? lk.reduce_buf : nullptr);
+          if (!lk.rccl)  // TCP: stage the whole message to pinned host
+            HIP_TRY(hipMemcpyAsync(lk.send_pin + 8, lk.send_buf, SA_ + P_,
+                                   hipMemcpyDeviceToHost, lk.s_send));
+        };
+        if (cfg_.use_graphs) {
+          if (!lk.g_quant) lk.g_quant = capture_seq(lk.s_send, qseq);
+          HIP_TRY(hipGraphLaunch(lk.g_quant, lk.s_send));
+        } else {
+          qseq();
+        }
         lagged_valid = true;
       } else {
         for (int t = 0; t < T_; ++t)
@@ -1028,12 +1075,20 @@ void Engine::apply_packet(Link& lk, const float* scales_host) {
     if (i != lk.idx && links_[i].provisioned) fwd[nf++] = links_[i].delta;
   if (gpu()) {
     HIP_TRY(hipSetDevice(cfg_.device));
-    if (!lk.rccl)  // TCP staging; RCCL already delivered into recv_buf
-      HIP_TRY(hipMemcpyAsync(lk.recv_buf, lk.recv_pin + 8, SA_ + P_,
-                             hipMemcpyHostToDevice, lk.s_recv));
-    hip_apply(cfg_.codec, lk.recv_buf + SA_, dtb_,
-              reinterpret_cast<float*>(lk.recv_buf), values_, fwd[0], fwd[1],
-              nullptr, lk.s_recv);
+    auto aseq = [&] {
+      if (!lk.rccl)  // TCP staging; RCCL already delivered into recv_buf
+        HIP_TRY(hipMemcpyAsync(lk.recv_buf, lk.recv_pin + 8, SA_ + P_,
+                               hipMemcpyHostToDevice, lk.s_recv));
+      hip_apply(cfg_.codec, lk.recv_buf + SA_, dtb_,
+                reinterpret_cast<float*>(lk.recv_buf), values_, fwd[0], fwd[1],
+                nullptr, lk.s_recv);
+    };
+    if (cfg_.use_graphs) {
+      if (!lk.g_apply) lk.g_apply = capture_seq(lk.s_recv, aseq);
+      HIP_TRY(hipGraphLaunch(lk.g_apply, lk.s_recv));
+    } else {
+      aseq();
+    }
     HIP_TRY(hipStreamSynchronize(lk.s_recv));
   } else {
     for (int t = 0; t < T_; ++t) {

@@ -51,6 +51,13 @@ struct Link {
   void* rccl_link = nullptr;   // opaque RcclLink*
   std::thread t_ctrl;          // TCP control reader when data plane is RCCL
   std::atomic<bool> abort{false};
+  // captured per-round sequences (GPU): one hipGraphLaunch replaces the
+  // 2-4 launches of the scale / quantize+stage / apply phases
+  hipGraphExec_t g_scale_lagged = nullptr;  // finalize+memset+scales D2H
+  hipGraphExec_t g_scale_exact = nullptr;   // reduce+finalize(+memset)+D2H
+  hipGraphExec_t g_quant = nullptr;         // quantize (+payload D2H for TCP)
+  hipGraphExec_t g_apply = nullptr;         // (H2D for TCP) + apply
+  std::vector<float> scales_host;           // persistent D2H target
 };
 
 class Engine {

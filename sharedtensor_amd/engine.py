@@ -41,7 +41,8 @@ class _SharedBase:
                  bw_limit: float = 0.0, expected_children: int = 2,
                  provision_up: bool = True, explicit_parent: str = "",
                  listen_port: int = 0, join_timeout_s: float = 60.0,
-                 rms_sample_stride: int = 1, lagged_scale: bool = False):
+                 rms_sample_stride: int = 1, lagged_scale: bool = False,
+                 use_graphs: bool = True):
         if codec not in CODECS:
             raise ValueError(f"codec must be one of {list(CODECS)}")
         self.device = torch.device(device)
@@ -63,6 +64,7 @@ class _SharedBase:
         cfg.join_timeout_s = join_timeout_s
         cfg.rms_sample_stride = rms_sample_stride
         cfg.lagged_scale = bool(lagged_scale)
+        cfg.use_graphs = bool(use_graphs)
         self.codec = codec
         self.n = int(sum(sizes))
         self._cfg = cfg
