@@ -94,6 +94,7 @@ def run_train(args, rank, world, device):
         param_dtype=torch.bfloat16 if use_bf16_params else torch.float32,
         codec=args.codec, use_rccl=not args.no_rccl,
         lagged_scale=(not args.exact_scale) and device.type == "cuda",
+        use_graphs=not args.no_graphs,
         snapshot_join=True)
 
     B, T = args.batch, cfg.block_size
@@ -193,6 +194,7 @@ def run_table(args, rank, world, device):
         "127.0.0.1", port_base, {n: p.data for n, p in model.named_parameters()},
         codec=args.codec,
         lagged_scale=(not args.exact_scale) and device.type == "cuda",
+        use_graphs=not args.no_graphs,
         use_rccl=not args.no_rccl, expected_children=nchild if world > 1 else 0,
         provision_up=rank > 0,
         explicit_parent=f"127.0.0.1:{port_base + tree_parent(rank)}" if rank else "",
@@ -262,6 +264,7 @@ def run_paramsync(args, rank, world, device):
     sh = SharedFlat(
         "127.0.0.1", port_base, [n], device=device, codec=args.codec,
         lagged_scale=(not args.exact_scale) and device.type == "cuda",
+        use_graphs=not args.no_graphs,
         use_rccl=not args.no_rccl, expected_children=nchild if world > 1 else 0,
         provision_up=rank > 0,
         explicit_parent=f"127.0.0.1:{port_base + tree_parent(rank)}" if rank else "",
@@ -335,6 +338,8 @@ def main():
     ap.add_argument("--no-rccl", action="store_true")
     ap.add_argument("--fp32-params", action="store_true",
                     help="compute on fp32 replica views (default: bf16 shadow)")
+    ap.add_argument("--no-graphs", action="store_true",
+                    help="disable hipGraph capture of sync rounds")
     ap.add_argument("--exact-scale", action="store_true",
                     help="per-round exact scale reduce (default: lagged, fused into quantize)")
     ap.add_argument("--device", default="auto")
