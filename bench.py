@@ -94,7 +94,8 @@ def run_train(args, rank, world, device):
         param_dtype=torch.bfloat16 if use_bf16_params else torch.float32,
         codec=args.codec, use_rccl=not args.no_rccl,
         lagged_scale=(not args.exact_scale) and device.type == "cuda",
-        use_graphs=not args.no_graphs,
+        use_graphs=args.graphs,
+        delta_dtype=torch.bfloat16 if args.bf16_deltas else torch.float32,
         snapshot_join=True)
 
     B, T = args.batch, cfg.block_size
@@ -194,7 +195,7 @@ def run_table(args, rank, world, device):
         "127.0.0.1", port_base, {n: p.data for n, p in model.named_parameters()},
         codec=args.codec,
         lagged_scale=(not args.exact_scale) and device.type == "cuda",
-        use_graphs=not args.no_graphs,
+        use_graphs=args.graphs,
         use_rccl=not args.no_rccl, expected_children=nchild if world > 1 else 0,
         provision_up=rank > 0,
         explicit_parent=f"127.0.0.1:{port_base + tree_parent(rank)}" if rank else "",
@@ -264,7 +265,8 @@ def run_paramsync(args, rank, world, device):
     sh = SharedFlat(
         "127.0.0.1", port_base, [n], device=device, codec=args.codec,
         lagged_scale=(not args.exact_scale) and device.type == "cuda",
-        use_graphs=not args.no_graphs,
+        use_graphs=args.graphs,
+        delta_dtype=torch.bfloat16 if args.bf16_deltas else torch.float32,
         use_rccl=not args.no_rccl, expected_children=nchild if world > 1 else 0,
         provision_up=rank > 0,
         explicit_parent=f"127.0.0.1:{port_base + tree_parent(rank)}" if rank else "",
@@ -338,8 +340,12 @@ def main():
     ap.add_argument("--no-rccl", action="store_true")
     ap.add_argument("--fp32-params", action="store_true",
                     help="compute on fp32 replica views (default: bf16 shadow)")
-    ap.add_argument("--no-graphs", action="store_true",
-                    help="disable hipGraph capture of sync rounds")
+    ap.add_argument("--bf16-deltas", action="store_true",
+                    help="store link residual deltas in bf16 (halves their "
+                         "HBM footprint; for 100GB-scale tensors)")
+    ap.add_argument("--graphs", action="store_true",
+                    help="capture sync rounds into hipGraphs (measured "
+                         "slower on MI355X; see profiles/README.md)")
     ap.add_argument("--exact-scale", action="store_true",
                     help="per-round exact scale reduce (default: lagged, fused into quantize)")
     ap.add_argument("--device", default="auto")

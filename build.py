@@ -97,6 +97,13 @@ def build(force=False, debug=False, verbose=True):
     if verbose:
         print("[build.py]", " ".join(link))
     subprocess.check_call(link)
+    # guard against silently-unresolved engine symbols (python extensions
+    # link with undefined symbols allowed; a stale signature would only
+    # crash at first GPU call otherwise)
+    nm = subprocess.run(["nm", "-C", "-u", OUT], capture_output=True, text=True)
+    bad = [l for l in nm.stdout.splitlines() if "shamd::" in l]
+    if bad:
+        raise RuntimeError("unresolved engine symbols:\n" + "\n".join(bad))
     with open(stamp, "w") as fh:
         fh.write(digest)
     if verbose:

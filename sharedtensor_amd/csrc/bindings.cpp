@@ -46,10 +46,11 @@ struct DevCodec {
   DevTable tb{};
   Codec c;
   int device;
+  bool dbf16;
   void* reduce = nullptr;
 
-  DevCodec(int codec, std::vector<int64_t> sizes, int dev)
-      : c(static_cast<Codec>(codec)), device(dev) {
+  DevCodec(int codec, std::vector<int64_t> sizes, int dev, bool delta_bf16)
+      : c(static_cast<Codec>(codec)), device(dev), dbf16(delta_bf16) {
     std::vector<int64_t> offs(sizes.size() + 1), poffs(sizes.size() + 1);
     offs[0] = poffs[0] = 0;
     for (size_t t = 0; t < sizes.size(); ++t) {
@@ -78,13 +79,13 @@ struct DevCodec {
   }
   void reduce_scales(uintptr_t delta, uintptr_t scales_dev, int stride,
                      uintptr_t stream) {
-    hip_reduce_scales(c, reinterpret_cast<const float*>(delta), tb, reduce,
-                      reinterpret_cast<float*>(scales_dev), stride,
+    hip_reduce_scales(c, reinterpret_cast<const void*>(delta), dbf16, tb,
+                      reduce, reinterpret_cast<float*>(scales_dev), stride,
                       reinterpret_cast<hipStream_t>(stream));
   }
   void quantize(uintptr_t delta, uintptr_t scales_dev, uintptr_t payload,
                 uintptr_t stream, uintptr_t stats = 0) {
-    hip_quantize(c, reinterpret_cast<float*>(delta), tb,
+    hip_quantize(c, reinterpret_cast<void*>(delta), dbf16, tb,
                  reinterpret_cast<const float*>(scales_dev),
                  reinterpret_cast<uint8_t*>(payload),
                  reinterpret_cast<hipStream_t>(stream),
@@ -95,35 +96,43 @@ struct DevCodec {
                         reinterpret_cast<float*>(scales_dev), 1,
                         reinterpret_cast<hipStream_t>(stream));
   }
+  // dsts[0] = fp32 values (or 0); dsts[1..2] = delta-typed forwards
   void apply(uintptr_t payload, uintptr_t scales_dev,
              std::vector<uintptr_t> dsts, uintptr_t stream) {
-    float* d[4] = {nullptr, nullptr, nullptr, nullptr};
-    for (size_t i = 0; i < dsts.size() && i < 4; ++i)
-      d[i] = reinterpret_cast<float*>(dsts[i]);
+    uintptr_t d[3] = {0, 0, 0};
+    for (size_t i = 0; i < dsts.size() && i < 3; ++i) d[i] = dsts[i];
     hip_apply(c, reinterpret_cast<const uint8_t*>(payload), tb,
-              reinterpret_cast<const float*>(scales_dev), d[0], d[1], d[2],
-              d[3], reinterpret_cast<hipStream_t>(stream));
+              reinterpret_cast<const float*>(scales_dev),
+              reinterpret_cast<float*>(d[0]), reinterpret_cast<void*>(d[1]),
+              reinterpret_cast<void*>(d[2]), dbf16,
+              reinterpret_cast<hipStream_t>(stream));
   }
 };
 
+// dsts[0] = fp32 values (or 0); dsts[1..3] = delta-typed
 void py_gpu_add_scatter(uintptr_t src, int64_t n, double alpha,
-                        std::vector<uintptr_t> dsts, uintptr_t stream) {
-  float* d[4] = {nullptr, nullptr, nullptr, nullptr};
-  for (size_t i = 0; i < dsts.size() && i < 4; ++i)
-    d[i] = reinterpret_cast<float*>(dsts[i]);
+                        std::vector<uintptr_t> dsts, uintptr_t stream,
+                        bool delta_bf16) {
+  uintptr_t d[4] = {0, 0, 0, 0};
+  for (size_t i = 0; i < dsts.size() && i < 4; ++i) d[i] = dsts[i];
   hip_add_scatter(reinterpret_cast<const float*>(src), n,
-                  static_cast<float>(alpha), d[0], d[1], d[2], d[3],
+                  static_cast<float>(alpha), reinterpret_cast<float*>(d[0]),
+                  reinterpret_cast<void*>(d[1]), reinterpret_cast<void*>(d[2]),
+                  reinterpret_cast<void*>(d[3]), delta_bf16,
                   reinterpret_cast<hipStream_t>(stream));
 }
 
+// dsts[0] = fp32 values (or 0); dsts[1..3] = delta-typed
 void py_gpu_fused_sgd(uintptr_t mom, uintptr_t grad, double lr, double mu,
-                      int64_t n, std::vector<uintptr_t> dsts, uintptr_t stream) {
-  float* d[4] = {nullptr, nullptr, nullptr, nullptr};
-  for (size_t i = 0; i < dsts.size() && i < 4; ++i)
-    d[i] = reinterpret_cast<float*>(dsts[i]);
+                      int64_t n, std::vector<uintptr_t> dsts, uintptr_t stream,
+                      bool delta_bf16) {
+  uintptr_t d[4] = {0, 0, 0, 0};
+  for (size_t i = 0; i < dsts.size() && i < 4; ++i) d[i] = dsts[i];
   hip_fused_sgd(reinterpret_cast<float*>(mom),
                 reinterpret_cast<const float*>(grad), static_cast<float>(lr),
-                static_cast<float>(mu), n, d[0], d[1], d[2], d[3],
+                static_cast<float>(mu), n, reinterpret_cast<float*>(d[0]),
+                reinterpret_cast<void*>(d[1]), reinterpret_cast<void*>(d[2]),
+                reinterpret_cast<void*>(d[3]), delta_bf16,
                 reinterpret_cast<hipStream_t>(stream));
 }
 
@@ -152,6 +161,7 @@ PYBIND11_MODULE(_core, m) {
       .def_readwrite("join_timeout_s", &Config::join_timeout_s)
       .def_readwrite("rms_sample_stride", &Config::rms_sample_stride)
       .def_readwrite("lagged_scale", &Config::lagged_scale)
+      .def_readwrite("delta_bf16", &Config::delta_bf16)
       .def_readwrite("use_graphs", &Config::use_graphs);
 
   py::class_<Engine>(m, "Engine")
@@ -195,15 +205,20 @@ PYBIND11_MODULE(_core, m) {
       });
 
   py::class_<DevCodec>(m, "DevCodec")
-      .def(py::init<int, std::vector<int64_t>, int>())
+      .def(py::init<int, std::vector<int64_t>, int, bool>(), py::arg("codec"),
+           py::arg("sizes"), py::arg("device"), py::arg("delta_bf16") = false)
       .def("reduce_scales", &DevCodec::reduce_scales)
       .def("quantize", &DevCodec::quantize, py::arg("delta"),
            py::arg("scales"), py::arg("payload"), py::arg("stream"),
            py::arg("stats") = 0)
       .def("finalize_scales", &DevCodec::finalize_scales)
       .def("apply", &DevCodec::apply);
-  m.def("gpu_add_scatter", &py_gpu_add_scatter);
-  m.def("gpu_fused_sgd", &py_gpu_fused_sgd);
+  m.def("gpu_add_scatter", &py_gpu_add_scatter, py::arg("src"), py::arg("n"),
+        py::arg("alpha"), py::arg("dsts"), py::arg("stream"),
+        py::arg("delta_bf16") = false);
+  m.def("gpu_fused_sgd", &py_gpu_fused_sgd, py::arg("mom"), py::arg("grad"),
+        py::arg("lr"), py::arg("mu"), py::arg("n"), py::arg("dsts"),
+        py::arg("stream"), py::arg("delta_bf16") = false);
   m.def("rccl_self_test", &rccl_self_test,
         py::call_guard<py::gil_scoped_release>());
 

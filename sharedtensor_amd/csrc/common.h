@@ -66,9 +66,14 @@ struct Config {
                               // quantize kernel (next round's scale from this
                               // round's post-quantize residual) — removes the
                               // reduce pass from the steady state
-  bool use_graphs = true;     // GPU: capture the per-round kernel+copy
-                              // sequences into hipGraphs (one replay instead
-                              // of 2-4 launches per round)
+  bool delta_bf16 = false;    // GPU: store per-link residual deltas in bf16
+                              // (halves their HBM footprint; debit quanta
+                              // are bf16-exact, only the remainder rounds)
+  bool use_graphs = false;    // GPU: capture the per-round kernel+copy
+                              // sequences into hipGraphs.  Measured NET
+                              // NEGATIVE on MI355X/ROCm 7.2 (replay floor
+                              // ~10-16us vs ~3.5us per plain launch; rounds
+                              // are 2-4 ops) — kept as an option only.
 };
 
 struct LinkStatsSnap {

@@ -161,6 +161,8 @@ Engine::Engine(Config cfg) : cfg_(std::move(cfg)) {
   pe_ = poffs_[T_];
   SA_ = scales_area(cfg_);
   P_ = msg_bytes(cfg_) - SA_;
+  if (cfg_.delta_bf16 && cfg_.device < 0)
+    throw std::runtime_error("delta_bf16 requires a GPU engine");
   hostid_ = compute_hostid();
   for (int i = 0; i < 3; ++i) links_[i].idx = i;
 }
@@ -178,7 +180,7 @@ void Engine::set_link_buffers(int link, uintptr_t delta, uintptr_t send_buf,
                               uintptr_t recv_buf, uintptr_t send_pin,
                               uintptr_t recv_pin) {
   Link& lk = links_[link];
-  lk.delta = reinterpret_cast<float*>(delta);
+  lk.delta = reinterpret_cast<void*>(delta);
   lk.send_buf = reinterpret_cast<uint8_t*>(send_buf);
   lk.recv_buf = reinterpret_cast<uint8_t*>(recv_buf);
   lk.send_pin = reinterpret_cast<uint8_t*>(send_pin);
@@ -420,19 +422,19 @@ void Engine::handshake_as_child(int fd, bool rejoin) {
     // (slot delta == values) is rebuilt by the same adds.
     zero_buf(values_, n_);
     recv_snapshot(fd);  // += S into values + child slots
-    float* fwd0 = links_[LK_LEFT].provisioned ? links_[LK_LEFT].delta : nullptr;
-    float* fwd1 = links_[LK_RIGHT].provisioned ? links_[LK_RIGHT].delta : nullptr;
+    void* fwd0 = links_[LK_LEFT].provisioned ? links_[LK_LEFT].delta : nullptr;
+    void* fwd1 = links_[LK_RIGHT].provisioned ? links_[LK_RIGHT].delta : nullptr;
     if (gpu()) {
-      hip_add_scatter(up.delta, n_, 1.0f, values_, fwd0, fwd1, nullptr,
-                      up.s_recv);
+      hip_add_delta_scatter(up.delta, cfg_.delta_bf16, n_, values_, fwd0,
+                            fwd1, up.s_recv);
       HIP_TRY(hipStreamSynchronize(up.s_recv));
     } else {
       for (int64_t i = 0; i < n_; ++i) {
-        float v = atomic_load_f32(up.delta + i);
+        float v = atomic_load_f32(fdelta(up.delta) + i);
         if (v == 0.0f) continue;
         atomic_add_f32(values_ + i, v);
-        if (fwd0) atomic_add_f32(fwd0 + i, v);
-        if (fwd1) atomic_add_f32(fwd1 + i, v);
+        if (fwd0) atomic_add_f32(fdelta(fwd0) + i, v);
+        if (fwd1) atomic_add_f32(fdelta(fwd1) + i, v);
       }
     }
   } else if (ah.flags & ACC_SNAPSHOT) {
@@ -449,6 +451,24 @@ void Engine::zero_buf(float* p, int64_t n) {
   } else {
     std::memset(p, 0, n * 4);
   }
+}
+
+void Engine::zero_delta(void* p) {
+  int64_t bytes = n_ * (cfg_.delta_bf16 ? 2 : 4);
+  if (gpu()) {
+    HIP_TRY(hipMemset(p, 0, bytes));
+  } else {
+    std::memset(p, 0, bytes);
+  }
+}
+
+void* Engine::doff(void* delta, int64_t off) const {
+  return static_cast<char*>(delta) + off * (cfg_.delta_bf16 ? 2 : 4);
+}
+
+float* Engine::fdelta(void* p) const {
+  // CPU engines always keep fp32 residuals (validated in the constructor)
+  return static_cast<float*>(p);
 }
 
 void Engine::drop_children() {
@@ -475,7 +495,7 @@ void Engine::drop_children() {
     lk.abort.store(false);
     lk.error.clear();
     if (gpu()) destroy_link_graphs(lk);  // transport may change on reuse
-    if (lk.provisioned) zero_buf(lk.delta, n_);
+    if (lk.provisioned) zero_delta(lk.delta);
     lk.state.store(L_FREE);
   }
 }
@@ -548,20 +568,20 @@ void Engine::reconnect_loop() try {
           // restore the unconnected-slot invariant (slot delta == values,
           // sharedtensor.c:379-381 semantics): future children must receive
           // the full inherited state
-          float* fwd0 = links_[LK_LEFT].provisioned ? links_[LK_LEFT].delta : nullptr;
-          float* fwd1 = links_[LK_RIGHT].provisioned ? links_[LK_RIGHT].delta : nullptr;
+          void* fwd0 = links_[LK_LEFT].provisioned ? links_[LK_LEFT].delta : nullptr;
+          void* fwd1 = links_[LK_RIGHT].provisioned ? links_[LK_RIGHT].delta : nullptr;
           if (fwd0 || fwd1) {
             if (gpu()) {
               HIP_TRY(hipSetDevice(cfg_.device));
-              hip_add_scatter(values_, n_, 1.0f, fwd0, fwd1, nullptr, nullptr,
-                              nullptr);
+              hip_add_scatter(values_, n_, 1.0f, nullptr, fwd0, fwd1, nullptr,
+                              cfg_.delta_bf16, nullptr);
               HIP_TRY(hipStreamSynchronize(nullptr));
             } else {
               for (int64_t i = 0; i < n_; ++i) {
                 float v = atomic_load_f32(values_ + i);
                 if (v == 0.0f) continue;
-                if (fwd0) atomic_add_f32(fwd0 + i, v);
-                if (fwd1) atomic_add_f32(fwd1 + i, v);
+                if (fwd0) atomic_add_f32(fdelta(fwd0) + i, v);
+                if (fwd1) atomic_add_f32(fdelta(fwd1) + i, v);
               }
             }
           }
@@ -860,7 +880,8 @@ void Engine::send_snapshot(Link& lk) {
       HIP_TRY(hipMemcpyAsync(lk.send_buf, lk.send_pin, ce * 4,
                              hipMemcpyHostToDevice, lk.s_send));
       hip_add_scatter(reinterpret_cast<float*>(lk.send_buf), ce, -1.0f,
-                      lk.delta + off, nullptr, nullptr, nullptr, lk.s_send);
+                      nullptr, doff(lk.delta, off), nullptr, nullptr,
+                      cfg_.delta_bf16, lk.s_send);
       HIP_TRY(hipStreamSynchronize(lk.s_send));
     } else {
       for (int64_t i = 0; i < ce; ++i)
@@ -869,7 +890,8 @@ void Engine::send_snapshot(Link& lk) {
         throw std::runtime_error("tcp write failed");
       const float* snap = reinterpret_cast<const float*>(tmp.data());
       for (int64_t i = 0; i < ce; ++i)
-        if (snap[i] != 0.0f) atomic_add_f32(lk.delta + off + i, -snap[i]);
+        if (snap[i] != 0.0f)
+          atomic_add_f32(fdelta(lk.delta) + off + i, -snap[i]);
     }
     lk.bytes_sent += ce * 4;
   }
@@ -881,7 +903,7 @@ void Engine::recv_snapshot(int fd) {
   const int64_t chunk_elems = std::max<int64_t>(chunk_bytes / 4, 1);
   std::vector<uint8_t> tmp;
   if (!gpu()) tmp.resize(static_cast<size_t>(chunk_elems) * 4);
-  float* fwd[2] = {
+  void* fwd[2] = {
       links_[LK_LEFT].provisioned ? links_[LK_LEFT].delta : nullptr,
       links_[LK_RIGHT].provisioned ? links_[LK_RIGHT].delta : nullptr};
   for (int64_t off = 0; off < n_; off += chunk_elems) {
@@ -895,8 +917,9 @@ void Engine::recv_snapshot(int fd) {
       // values += snap; future children's deltas += snap (join-state
       // forwarding, the GPU analog of sharedtensor.c:379-381)
       hip_add_scatter(reinterpret_cast<float*>(up.recv_buf), ce, 1.0f,
-                      values_ + off, fwd[0] ? fwd[0] + off : nullptr,
-                      fwd[1] ? fwd[1] + off : nullptr, nullptr, up.s_recv);
+                      values_ + off, fwd[0] ? doff(fwd[0], off) : nullptr,
+                      fwd[1] ? doff(fwd[1], off) : nullptr, nullptr,
+                      cfg_.delta_bf16, up.s_recv);
       HIP_TRY(hipStreamSynchronize(up.s_recv));
     } else {
       if (!io_read(fd, tmp.data(), ce * 4))
@@ -906,8 +929,8 @@ void Engine::recv_snapshot(int fd) {
         float v = snap[i];
         if (v == 0.0f) continue;
         atomic_add_f32(values_ + off + i, v);
-        if (fwd[0]) atomic_add_f32(fwd[0] + off + i, v);
-        if (fwd[1]) atomic_add_f32(fwd[1] + off + i, v);
+        if (fwd[0]) atomic_add_f32(fdelta(fwd[0]) + off + i, v);
+        if (fwd[1]) atomic_add_f32(fdelta(fwd[1]) + off + i, v);
       }
     }
     up.bytes_recv += ce * 4;
@@ -931,8 +954,9 @@ void Engine::compute_scales(Link& lk, float* scales_host, bool lagged_valid) {
                             /*stride=*/1, lk.s_send);
         HIP_TRY(hipMemsetAsync(lk.reduce_buf, 0, 8 * T_, lk.s_send));
       } else {
-        hip_reduce_scales(cfg_.codec, lk.delta, dtb_, lk.reduce_buf,
-                          scales_dev, cfg_.rms_sample_stride, lk.s_send);
+        hip_reduce_scales(cfg_.codec, lk.delta, cfg_.delta_bf16, dtb_,
+                          lk.reduce_buf, scales_dev, cfg_.rms_sample_stride,
+                          lk.s_send);
         if (cfg_.lagged_scale)
           HIP_TRY(hipMemsetAsync(lk.reduce_buf, 0, 8 * T_, lk.s_send));
       }
@@ -950,7 +974,7 @@ void Engine::compute_scales(Link& lk, float* scales_host, bool lagged_valid) {
     std::memcpy(scales_host, lk.scales_host.data(), 4 * T_);
   } else {
     for (int t = 0; t < T_; ++t)
-      scales_host[t] = cpu_compute_scale(cfg_.codec, lk.delta + offs_[t],
+      scales_host[t] = cpu_compute_scale(cfg_.codec, fdelta(lk.delta) + offs_[t],
                                          cfg_.sizes[t], cfg_.rms_sample_stride);
     std::memcpy(lk.send_pin + 8, scales_host, 4 * T_);
   }
@@ -1021,7 +1045,7 @@ void Engine::send_loop(Link& lk) {
     try {
       if (gpu()) {
         auto qseq = [&] {
-          hip_quantize(cfg_.codec, lk.delta, dtb_,
+          hip_quantize(cfg_.codec, lk.delta, cfg_.delta_bf16, dtb_,
                        reinterpret_cast<float*>(lk.send_buf), lk.send_buf + SA_,
                        lk.s_send, cfg_.lagged_scale ? lk.reduce_buf : nullptr);
           if (!lk.rccl)  // TCP: stage the whole message to pinned host
@@ -1037,7 +1061,7 @@ void Engine::send_loop(Link& lk) {
         lagged_valid = true;
       } else {
         for (int t = 0; t < T_; ++t)
-          cpu_quantize(cfg_.codec, lk.delta + offs_[t], cfg_.sizes[t], scales[t],
+          cpu_quantize(cfg_.codec, fdelta(lk.delta) + offs_[t], cfg_.sizes[t], scales[t],
                        lk.send_pin + 8 + SA_ +
                            (poffs_[t] / 64) * (payload_bytes(cfg_.codec, 64)));
       }
@@ -1069,7 +1093,7 @@ void Engine::send_loop(Link& lk) {
 void Engine::apply_packet(Link& lk, const float* scales_host) {
   // destinations: local replica + gossip-forward into the other links'
   // delta buffers, excluding the source (sharedtensor.c:124-127)
-  float* fwd[2] = {nullptr, nullptr};
+  void* fwd[2] = {nullptr, nullptr};
   int nf = 0;
   for (int i = 0; i < 3; ++i)
     if (i != lk.idx && links_[i].provisioned) fwd[nf++] = links_[i].delta;
@@ -1081,7 +1105,7 @@ void Engine::apply_packet(Link& lk, const float* scales_host) {
                                hipMemcpyHostToDevice, lk.s_recv));
       hip_apply(cfg_.codec, lk.recv_buf + SA_, dtb_,
                 reinterpret_cast<float*>(lk.recv_buf), values_, fwd[0], fwd[1],
-                nullptr, lk.s_recv);
+                cfg_.delta_bf16, lk.s_recv);
     };
     if (cfg_.use_graphs) {
       if (!lk.g_apply) lk.g_apply = capture_seq(lk.s_recv, aseq);
@@ -1093,8 +1117,8 @@ void Engine::apply_packet(Link& lk, const float* scales_host) {
   } else {
     for (int t = 0; t < T_; ++t) {
       float* dsts[3] = {values_ + offs_[t],
-                        fwd[0] ? fwd[0] + offs_[t] : nullptr,
-                        fwd[1] ? fwd[1] + offs_[t] : nullptr};
+                        fwd[0] ? fdelta(fwd[0]) + offs_[t] : nullptr,
+                        fwd[1] ? fdelta(fwd[1]) + offs_[t] : nullptr};
       int nd = 1 + (fwd[0] ? 1 : 0) + (fwd[1] ? 1 : 0);
       cpu_apply(cfg_.codec,
                 lk.recv_pin + 8 + SA_ +
@@ -1222,18 +1246,18 @@ void Engine::notify_dirty() { notify_all_dirty(); }
 void Engine::add_from(uintptr_t src, int64_t n, uintptr_t stream) {
   if (n != n_) throw std::runtime_error("add_from: size mismatch");
   const float* s = reinterpret_cast<const float*>(src);
-  float* d[3];
+  void* d[3];
   for (int i = 0; i < 3; ++i)
     d[i] = links_[i].provisioned ? links_[i].delta : nullptr;
   if (gpu()) {
-    hip_add_scatter(s, n_, 1.0f, values_, d[0], d[1], d[2],
+    hip_add_scatter(s, n_, 1.0f, values_, d[0], d[1], d[2], cfg_.delta_bf16,
                     reinterpret_cast<hipStream_t>(stream));
   } else {
     float* dsts[4];
     int nd = 0;
     dsts[nd++] = values_;
     for (int i = 0; i < 3; ++i)
-      if (d[i]) dsts[nd++] = d[i];
+      if (d[i]) dsts[nd++] = fdelta(d[i]);
     cpu_add_scatter(s, n_, dsts, nd);
   }
   notify_all_dirty();
@@ -1253,15 +1277,14 @@ void Engine::copy_to(uintptr_t dst, int64_t n, uintptr_t stream) {
 
 void Engine::fused_sgd(uintptr_t mom, uintptr_t grad, double lr,
                        double momentum, uintptr_t stream) {
-  float* d[3];
+  void* d[3];
   for (int i = 0; i < 3; ++i)
     d[i] = links_[i].provisioned ? links_[i].delta : nullptr;
   if (gpu()) {
-    // values is always a destination; pack it with the link deltas
     hip_fused_sgd(reinterpret_cast<float*>(mom),
                   reinterpret_cast<const float*>(grad), static_cast<float>(lr),
                   static_cast<float>(momentum), n_, values_, d[0], d[1], d[2],
-                  reinterpret_cast<hipStream_t>(stream));
+                  cfg_.delta_bf16, reinterpret_cast<hipStream_t>(stream));
   } else {
     float* m = reinterpret_cast<float*>(mom);
     const float* g = reinterpret_cast<const float*>(grad);
@@ -1272,7 +1295,7 @@ void Engine::fused_sgd(uintptr_t mom, uintptr_t grad, double lr,
       if (u == 0.0f) continue;
       atomic_add_f32(values_ + i, u);
       for (int k = 0; k < 3; ++k)
-        if (d[k]) atomic_add_f32(d[k] + i, u);
+        if (d[k]) atomic_add_f32(fdelta(d[k]) + i, u);
     }
   }
   notify_all_dirty();
@@ -1283,14 +1306,14 @@ void Engine::fused_sgd_bf16(uintptr_t mom, uintptr_t grad_bf16,
                             uintptr_t stream) {
   if (!gpu())
     throw std::runtime_error("fused_sgd_bf16 is a GPU-only path");
-  float* d[3];
+  void* d[3];
   for (int i = 0; i < 3; ++i)
     d[i] = links_[i].provisioned ? links_[i].delta : nullptr;
   hip_fused_sgd_bf16(reinterpret_cast<float*>(mom),
                      reinterpret_cast<const uint16_t*>(grad_bf16),
                      reinterpret_cast<uint16_t*>(shadow_bf16),
                      static_cast<float>(lr), static_cast<float>(momentum), n_,
-                     values_, d[0], d[1], d[2],
+                     values_, d[0], d[1], d[2], cfg_.delta_bf16,
                      reinterpret_cast<hipStream_t>(stream));
   notify_all_dirty();
 }

@@ -30,7 +30,7 @@ struct Link {
                        // because a joiner binds its listener to the local
                        // addr of its up-connection, sharedtensor.c:292-316)
   bool provisioned = false;  // buffers exist for this slot
-  float* delta = nullptr;            // fp32[n], device or host
+  void* delta = nullptr;             // residual, fp32[n] or bf16[n] (GPU)
   uint8_t* send_buf = nullptr;       // [scales 4T | pad | payload P]
   uint8_t* recv_buf = nullptr;
   uint8_t* send_pin = nullptr;       // pinned host: [hdr 8 | scales | payload]
@@ -133,6 +133,9 @@ class Engine {
   void drop_children();
   bool failover_master();
   void zero_buf(float* p, int64_t n);
+  void zero_delta(void* p);
+  void* doff(void* delta, int64_t off) const;  // typed element offset
+  float* fdelta(void* p) const;  // CPU engines: deltas are always fp32
   void become_master();
   void bind_listen(const sockaddr_in& addr, bool shared = true);
   void listen_loop();

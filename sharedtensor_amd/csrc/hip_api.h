@@ -28,17 +28,22 @@ struct DevTable {
   int64_t pe = 0;             // poffs[T]
 };
 
+// Residual delta buffers are fp32 (delta_bf16=false) or bf16
+// (delta_bf16=true, halves delta HBM for 100 GB-scale tensors; debit quanta
+// are exactly representable in bf16).  `delta`/`d1..d3` pointers are typed
+// accordingly (void* at this interface).
+
 // reduce_buf: device scratch, >= 8*T bytes (double sumsq for 1bit,
 // fp32 absmax for fp8/int4). scales_out: device fp32[T].
-void hip_reduce_scales(Codec c, const float* delta, const DevTable& tb,
-                       void* reduce_buf, float* scales_out, int sample_stride,
-                       hipStream_t s);
+void hip_reduce_scales(Codec c, const void* delta, bool delta_bf16,
+                       const DevTable& tb, void* reduce_buf, float* scales_out,
+                       int sample_stride, hipStream_t s);
 
 // delta is debited in place (atomic, lossless vs concurrent adds);
 // payload receives the packed bytes for the whole padded space.
 // stats_out (nullable, 8*T bytes): accumulate post-quantize residual
 // statistics for lagged-scale mode (zero it before the call).
-void hip_quantize(Codec c, float* delta, const DevTable& tb,
+void hip_quantize(Codec c, void* delta, bool delta_bf16, const DevTable& tb,
                   const float* scales_dev, uint8_t* payload, hipStream_t s,
                   void* stats_out = nullptr);
 
@@ -47,28 +52,35 @@ void hip_quantize(Codec c, float* delta, const DevTable& tb,
 void hip_finalize_scales(Codec c, const DevTable& tb, const void* reduce_buf,
                          float* scales_out, int sample_stride, hipStream_t s);
 
-// Decode payload and accumulate into up to 4 destinations (values plus the
-// other links' delta buffers, sharedtensor.c:106-127).
+// Decode payload and accumulate into the fp32 replica (values, nullable)
+// plus up to two gossip-forward residual buffers (sharedtensor.c:106-127).
 void hip_apply(Codec c, const uint8_t* payload, const DevTable& tb,
-               const float* scales_dev, float* d0, float* d1, float* d2,
-               float* d3, hipStream_t s);
+               const float* scales_dev, float* values, void* d1, void* d2,
+               bool delta_bf16, hipStream_t s);
 
-// dst_i += alpha * src over the flat unpadded space (addFromInternal,
-// sharedtensor.c:334-344; alpha=-1 implements the snapshot debit).
-void hip_add_scatter(const float* src, int64_t n, float alpha, float* d0,
-                     float* d1, float* d2, float* d3, hipStream_t s);
+// {values, d1..d3} += alpha * src over the flat unpadded space
+// (addFromInternal, sharedtensor.c:334-344; alpha=-1 implements the
+// snapshot debit).
+void hip_add_scatter(const float* src, int64_t n, float alpha, float* values,
+                     void* d1, void* d2, void* d3, bool delta_bf16,
+                     hipStream_t s);
+
+// {values, d1, d2} += src_delta (delta-typed source; rejoin reconciliation).
+void hip_add_delta_scatter(const void* src_delta, bool delta_bf16, int64_t n,
+                           float* values, void* d1, void* d2, hipStream_t s);
 
 // Fused SGD-momentum update feeding the shared tensor: m = mu*m + g;
 // u = -lr*m; {values, link deltas} += u.  One pass over HBM instead of four.
 void hip_fused_sgd(float* mom, const float* grad, float lr, float momentum,
-                   int64_t n, float* d0, float* d1, float* d2, float* d3,
-                   hipStream_t s);
+                   int64_t n, float* values, void* d1, void* d2, void* d3,
+                   bool delta_bf16, hipStream_t s);
 
 // Mixed-precision variant: bf16 grads in, fp32 master (values) updated,
 // bf16 shadow params refreshed (folding concurrent gossip via the
 // atomicAdd return), link deltas staged — all in one pass.
 void hip_fused_sgd_bf16(float* mom, const uint16_t* grad, uint16_t* shadow,
                         float lr, float momentum, int64_t n, float* values,
-                        float* d1, float* d2, float* d3, hipStream_t s);
+                        void* d1, void* d2, void* d3, bool delta_bf16,
+                        hipStream_t s);
 
 }  // namespace shamd

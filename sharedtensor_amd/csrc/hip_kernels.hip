@@ -3,19 +3,29 @@
 // These replace the reference's single-threaded CPU loops
 // (/root/reference/src/sharedtensor.c:106-111, 156-174, 334-344) with
 // HBM-bandwidth-bound GPU passes:
-//   * k_reduce_*   — per-tensor RMS / absmax via 64-lane __shfl reduction +
-//                    one atomic per wavefront
-//   * k_quant_1bit — fused sign-quantize + error-feedback debit + __ballot
-//                    bit-pack (one uint64 per wave, LSB-first layout
-//                    byte-identical to the reference wire format)
-//   * k_apply_*    — fused decode + multi-destination atomic scatter
-//                    (values + gossip-forward buffers in one pass)
-//   * k_fused_sgd  — optimizer update fused with the 4-way delta scatter
+//   * k_reduce_*   — per-tensor RMS / absmax: per-lane register accumulation
+//                    across grid-stride iterations, one 64-lane __shfl tree +
+//                    one atomic per wave per tensor-run (naive per-iteration
+//                    atomics measured 21 GB/s — serialized f64 atomics)
+//   * k_quant_*    — fused quantize + error-feedback debit + pack; 1-bit uses
+//                    __ballot (one uint64 per wave, LSB-first layout
+//                    byte-identical to the reference wire format); optional
+//                    fused next-round scale statistics (lagged-scale mode)
+//   * k_apply      — fused decode + multi-destination atomic scatter
+//                    (replica + gossip-forward buffers in one pass)
+//   * k_add_scatter / k_fused_sgd* — addFromInternal and the optimizer update
+//                    fused with the 4-way delta staging
 //
-// All element loops are grid-stride with 256-thread blocks (4 waves); the
-// kernels are HBM-bound, so the win is minimizing passes over memory, not
-// MFMA work.  Wavefront size is 64 (CDNA4): __ballot returns uint64_t and a
-// wave maps exactly onto one packed word.
+// Residual delta buffers are fp32 by default or bf16 (DeltaBF16 policy) to
+// halve their HBM footprint for 100 GB-scale tensors: the per-round debit
+// quanta (+-scale, q*e4m3*pow2, int4*pow2) are exactly representable in
+// bf16, so only the accumulated remainder rounds.  bf16 accumulation uses
+// the hardware packed atomic (global_atomic_pk_add_bf16) on element pairs —
+// all padded regions are even-sized and 4-byte aligned.
+//
+// All element loops are grid-stride with 256-thread blocks (4 waves);
+// wavefront size is 64 (CDNA4).
+#include <hip/hip_bf16.h>
 #include <hip/hip_runtime.h>
 
 #include <stdexcept>
@@ -44,8 +54,48 @@ static inline int grid_for(int64_t n) {
 
 // ---------------------------------------------------------------- helpers
 
+__device__ __forceinline__ float bf16_to_f32(uint16_t u) {
+  return __uint_as_float(static_cast<uint32_t>(u) << 16);
+}
+
+__device__ __forceinline__ uint16_t f32_to_bf16(float f) {
+  uint32_t u = __float_as_uint(f);
+  if ((u & 0x7FFFFFFFu) > 0x7F800000u) return 0x7FC0;  // NaN
+  u += 0x7FFFu + ((u >> 16) & 1u);  // round-to-nearest-even
+  return static_cast<uint16_t>(u >> 16);
+}
+
+// Residual-delta storage policies --------------------------------------
+
+struct DeltaF32 {
+  using T = float;
+  static __device__ __forceinline__ float load(const T* p, int64_t i) {
+    return __hip_atomic_load(p + i, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+  }
+  static __device__ __forceinline__ void atomic_add(T* p, int64_t i, float v) {
+    atomicAdd(p + i, v);
+  }
+};
+
+struct DeltaBF16 {
+  using T = uint16_t;
+  static __device__ __forceinline__ float load(const T* p, int64_t i) {
+    return bf16_to_f32(p[i]);
+  }
+  static __device__ __forceinline__ void atomic_add(T* p, int64_t i, float v) {
+    // hardware packed bf16 atomic on the containing (even, odd) pair; the
+    // neighbour lane receives +0.0 (identity)
+    auto* base = reinterpret_cast<__hip_bfloat162*>(p + (i & ~int64_t(1)));
+    __hip_bfloat16 bv = __float2bfloat16(v);
+    __hip_bfloat16 bz = __float2bfloat16(0.0f);
+    __hip_bfloat162 val = (i & 1) ? __hip_bfloat162(bz, bv)
+                                  : __hip_bfloat162(bv, bz);
+    unsafeAtomicAdd(base, val);
+  }
+};
+
 // Binary search: largest t with poffs[t] <= j.  T is small (1..few hundred)
-// and the array is L2/LDS-hot; for T==1 the caller's fast path skips this.
+// and the array is L2-hot; for T==1 the caller's fast path skips this.
 __device__ __forceinline__ int find_tensor(const int64_t* poffs, int T, int64_t j) {
   int lo = 0, hi = T - 1;
   while (lo < hi) {
@@ -109,19 +159,25 @@ __device__ __forceinline__ float d_pow2_ceil(double x) {
 
 // ------------------------------------------------------------- reductions
 
-// Sum of squares per tensor (1-bit codec).  Each lane accumulates into a
-// register across its grid-stride iterations (a wave never straddles tensors
-// within an iteration: poffs are 64-aligned, so the flush condition is
-// wave-uniform); on tensor change and at the end, one __shfl tree + a single
-// double atomic per wave.  This keeps the atomic count at ~waves, not
-// ~elements/64 — the naive per-iteration atomic version measured 21 GB/s on
-// MI355X (serialized f64 atomics on one address).
 __device__ __forceinline__ void wave_flush_sumsq(double acc, int t, double* sumsq) {
   for (int w = 32; w > 0; w >>= 1) acc += __shfl_down(acc, w, 64);
   if ((threadIdx.x & 63) == 0 && acc != 0.0) atomicAdd(&sumsq[t], acc);
 }
 
-__global__ void k_reduce_sumsq(const float* __restrict__ delta,
+__device__ __forceinline__ void wave_flush_max(float acc, int t, uint32_t* amax) {
+  for (int w = 32; w > 0; w >>= 1) {
+    float o = __shfl_down(acc, w, 64);
+    acc = o > acc ? o : acc;
+  }
+  if ((threadIdx.x & 63) == 0 && acc > 0.0f)
+    atomicMax(&amax[t], __float_as_uint(acc));
+}
+
+// Sum of squares per tensor (1-bit codec).  A wave never straddles tensors
+// within an iteration (poffs are 64-aligned), so the flush condition is
+// wave-uniform.
+template <typename DP>
+__global__ void k_reduce_sumsq(const typename DP::T* __restrict__ delta,
                                const int64_t* offs, const int64_t* poffs,
                                int T, int64_t pe, int stride, double* sumsq) {
   int64_t gstride = static_cast<int64_t>(gridDim.x) * blockDim.x;
@@ -137,25 +193,15 @@ __global__ void k_reduce_sumsq(const float* __restrict__ delta,
     int64_t L = j - poffs[t];
     int64_t sz = offs[t + 1] - offs[t];
     if (L < sz && (stride == 1 || (L % stride) == 0)) {
-      double f = static_cast<double>(delta[offs[t] + L]);
+      double f = static_cast<double>(DP::load(delta, offs[t] + L));
       acc += f * f;
     }
   }
   if (cur_t >= 0) wave_flush_sumsq(acc, cur_t, sumsq);
 }
 
-// absmax per tensor (fp8/int4 codecs); same register-accumulate structure,
-// float atomicMax via uint compare (valid for non-negative floats).
-__device__ __forceinline__ void wave_flush_max(float acc, int t, uint32_t* amax) {
-  for (int w = 32; w > 0; w >>= 1) {
-    float o = __shfl_down(acc, w, 64);
-    acc = o > acc ? o : acc;
-  }
-  if ((threadIdx.x & 63) == 0 && acc > 0.0f)
-    atomicMax(&amax[t], __float_as_uint(acc));
-}
-
-__global__ void k_reduce_absmax(const float* __restrict__ delta,
+template <typename DP>
+__global__ void k_reduce_absmax(const typename DP::T* __restrict__ delta,
                                 const int64_t* offs, const int64_t* poffs,
                                 int T, int64_t pe, uint32_t* amax) {
   int64_t gstride = static_cast<int64_t>(gridDim.x) * blockDim.x;
@@ -171,7 +217,7 @@ __global__ void k_reduce_absmax(const float* __restrict__ delta,
     int64_t L = j - poffs[t];
     int64_t sz = offs[t + 1] - offs[t];
     if (L < sz) {
-      float v = fabsf(delta[offs[t] + L]);
+      float v = fabsf(DP::load(delta, offs[t] + L));
       acc = v > acc ? v : acc;
     }
   }
@@ -198,18 +244,14 @@ __global__ void k_finalize_scales(Codec c, const void* reduce_buf,
 }
 
 // --------------------------------------------------------------- quantize
+// The debit is an atomic add of -sent so concurrent adds (training thread,
+// gossip forwards) landing between the read and the update are preserved —
+// the GPU version of the reference's benign-race contract made lossless.
 
-// Fused sign-quantize + error feedback + __ballot bit-pack.  The debit is an
-// atomicAdd of -sent so concurrent adds (training thread, gossip forwards)
-// landing between the read and the update are preserved — the GPU version of
-// the reference's benign-race contract made lossless.
-// stats_out (nullable): accumulate the POST-quantize residual's sum of
-// squares per tensor (lagged-scale mode: next round's scale comes from this
-// round's quantize, so the separate reduce pass disappears from the steady
-// state).  Same register-accumulate + wave-flush structure as k_reduce_*.
-__global__ void k_quant_1bit(float* __restrict__ delta, const int64_t* offs,
-                             const int64_t* poffs, int T, int64_t pe,
-                             const float* __restrict__ scales,
+template <typename DP>
+__global__ void k_quant_1bit(typename DP::T* __restrict__ delta,
+                             const int64_t* offs, const int64_t* poffs, int T,
+                             int64_t pe, const float* __restrict__ scales,
                              uint64_t* __restrict__ words, double* stats_out) {
   int64_t gstride = static_cast<int64_t>(gridDim.x) * blockDim.x;
   double acc = 0.0;
@@ -226,12 +268,11 @@ __global__ void k_quant_1bit(float* __restrict__ delta, const int64_t* offs,
     float s = scales[t];
     bool neg = false;  // bit value: 1 means -scale was sent
     if (L < sz) {
-      float* p = delta + offs[t] + L;
-      float v = __hip_atomic_load(reinterpret_cast<float*>(p), __ATOMIC_RELAXED,
-                                  __HIP_MEMORY_SCOPE_AGENT);
+      int64_t g = offs[t] + L;
+      float v = DP::load(delta, g);
       if (s != 0.0f) {
         neg = !(v > 0.0f);
-        atomicAdd(p, neg ? s : -s);
+        DP::atomic_add(delta, g, neg ? s : -s);
       }
       if (stats_out) {
         // fp32 subtraction first: the statistic must see the same rounded
@@ -246,9 +287,10 @@ __global__ void k_quant_1bit(float* __restrict__ delta, const int64_t* offs,
   if (stats_out && cur_t >= 0) wave_flush_sumsq(acc, cur_t, stats_out);
 }
 
-__global__ void k_quant_fp8(float* __restrict__ delta, const int64_t* offs,
-                            const int64_t* poffs, int T, int64_t pe,
-                            const float* __restrict__ scales,
+template <typename DP>
+__global__ void k_quant_fp8(typename DP::T* __restrict__ delta,
+                            const int64_t* offs, const int64_t* poffs, int T,
+                            int64_t pe, const float* __restrict__ scales,
                             uint8_t* __restrict__ payload, uint32_t* stats_out) {
   int64_t gstride = static_cast<int64_t>(gridDim.x) * blockDim.x;
   float amax = 0.0f;
@@ -265,13 +307,13 @@ __global__ void k_quant_fp8(float* __restrict__ delta, const int64_t* offs,
     float s = scales[t];
     uint8_t q = 0;
     if (L < sz) {
-      float* p = delta + offs[t] + L;
-      float v = __hip_atomic_load(p, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+      int64_t g = offs[t] + L;
+      float v = DP::load(delta, g);
       float sent = 0.0f;
       if (s != 0.0f) {
         q = d_f32_to_e4m3(v / s);
         sent = d_e4m3_to_f32(q) * s;
-        atomicAdd(p, -sent);
+        DP::atomic_add(delta, g, -sent);
       }
       if (stats_out) {
         float r = fabsf(v - sent);
@@ -283,9 +325,10 @@ __global__ void k_quant_fp8(float* __restrict__ delta, const int64_t* offs,
   if (stats_out && cur_t >= 0) wave_flush_max(amax, cur_t, stats_out);
 }
 
-__global__ void k_quant_int4(float* __restrict__ delta, const int64_t* offs,
-                             const int64_t* poffs, int T, int64_t pe,
-                             const float* __restrict__ scales,
+template <typename DP>
+__global__ void k_quant_int4(typename DP::T* __restrict__ delta,
+                             const int64_t* offs, const int64_t* poffs, int T,
+                             int64_t pe, const float* __restrict__ scales,
                              uint8_t* __restrict__ payload, uint32_t* stats_out) {
   // one thread per payload byte = 2 elements (both in the same tensor:
   // padded regions are 64-aligned, hence even).  nb is only a multiple of
@@ -312,8 +355,8 @@ __global__ void k_quant_int4(float* __restrict__ delta, const int64_t* offs,
     for (int k = 0; k < 2; ++k) {
       int64_t L = base + k;
       if (act && L < sz) {
-        float* p = delta + offs[t] + L;
-        float v = __hip_atomic_load(p, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+        int64_t g = offs[t] + L;
+        float v = DP::load(delta, g);
         float sent = 0.0f;
         if (s != 0.0f) {
           float r = nearbyintf(v / s);
@@ -321,7 +364,7 @@ __global__ void k_quant_int4(float* __restrict__ delta, const int64_t* offs,
           int8_t q = static_cast<int8_t>(r);
           byte |= static_cast<uint8_t>(q & 0xF) << (k * 4);
           sent = static_cast<float>(q) * s;
-          atomicAdd(p, -sent);
+          DP::atomic_add(delta, g, -sent);
         }
         if (stats_out) {
           float r2 = fabsf(v - sent);
@@ -335,12 +378,14 @@ __global__ void k_quant_int4(float* __restrict__ delta, const int64_t* offs,
 }
 
 // ------------------------------------------------------------------ apply
+// destinations: the fp32 replica (values) and up to two gossip-forward
+// residual buffers (delta-typed), all in one pass over the payload.
 
-template <int CODEC>
+template <int CODEC, typename DP>
 __global__ void k_apply(const uint8_t* __restrict__ payload,
                         const int64_t* offs, const int64_t* poffs, int T,
                         int64_t pe, const float* __restrict__ scales,
-                        float* d0, float* d1, float* d2, float* d3) {
+                        float* values, typename DP::T* d1, typename DP::T* d2) {
   int64_t gstride = static_cast<int64_t>(gridDim.x) * blockDim.x;
   for (int64_t j = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; j < pe; j += gstride) {
     int t = (T == 1) ? 0 : find_tensor(poffs, T, j);
@@ -362,48 +407,74 @@ __global__ void k_apply(const uint8_t* __restrict__ payload,
       if (v == 0.0f) continue;
     }
     int64_t g = offs[t] + L;
-    if (d0) atomicAdd(d0 + g, v);
-    if (d1) atomicAdd(d1 + g, v);
-    if (d2) atomicAdd(d2 + g, v);
-    if (d3) atomicAdd(d3 + g, v);
+    if (values) atomicAdd(values + g, v);
+    if (d1) DP::atomic_add(d1, g, v);
+    if (d2) DP::atomic_add(d2, g, v);
   }
 }
 
+template <typename DP>
 __global__ void k_add_scatter(const float* __restrict__ src, int64_t n,
-                              float alpha, float* d0, float* d1, float* d2,
-                              float* d3) {
+                              float alpha, float* values, typename DP::T* d1,
+                              typename DP::T* d2, typename DP::T* d3) {
   int64_t gstride = static_cast<int64_t>(gridDim.x) * blockDim.x;
   for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n; i += gstride) {
     float v = alpha * src[i];
     if (v == 0.0f) continue;
-    if (d0) atomicAdd(d0 + i, v);
-    if (d1) atomicAdd(d1 + i, v);
-    if (d2) atomicAdd(d2 + i, v);
-    if (d3) atomicAdd(d3 + i, v);
+    if (values) atomicAdd(values + i, v);
+    if (d1) DP::atomic_add(d1, i, v);
+    if (d2) DP::atomic_add(d2, i, v);
+    if (d3) DP::atomic_add(d3, i, v);
   }
 }
 
-__device__ __forceinline__ float bf16_to_f32(uint16_t u) {
-  return __uint_as_float(static_cast<uint32_t>(u) << 16);
+// src is a (possibly bf16) residual buffer: used by the rejoin
+// reconciliation to re-add the preserved unsent up-residual into the fresh
+// replica and child slots.
+template <typename DP>
+__global__ void k_add_delta_scatter(const typename DP::T* __restrict__ src,
+                                    int64_t n, float* values,
+                                    typename DP::T* d1, typename DP::T* d2) {
+  int64_t gstride = static_cast<int64_t>(gridDim.x) * blockDim.x;
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n; i += gstride) {
+    float v = DP::load(src, i);
+    if (v == 0.0f) continue;
+    if (values) atomicAdd(values + i, v);
+    if (d1) DP::atomic_add(d1, i, v);
+    if (d2) DP::atomic_add(d2, i, v);
+  }
 }
 
-__device__ __forceinline__ uint16_t f32_to_bf16(float f) {
-  uint32_t u = __float_as_uint(f);
-  if ((u & 0x7FFFFFFFu) > 0x7F800000u) return 0x7FC0;  // NaN
-  u += 0x7FFFu + ((u >> 16) & 1u);  // round-to-nearest-even
-  return static_cast<uint16_t>(u >> 16);
+template <typename DP>
+__global__ void k_fused_sgd(float* __restrict__ mom,
+                            const float* __restrict__ grad, float lr,
+                            float momentum, int64_t n, float* values,
+                            typename DP::T* d1, typename DP::T* d2,
+                            typename DP::T* d3) {
+  int64_t gstride = static_cast<int64_t>(gridDim.x) * blockDim.x;
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n; i += gstride) {
+    float m = momentum * mom[i] + grad[i];
+    mom[i] = m;
+    float u = -lr * m;
+    if (values) atomicAdd(values + i, u);
+    if (d1) DP::atomic_add(d1, i, u);
+    if (d2) DP::atomic_add(d2, i, u);
+    if (d3) DP::atomic_add(d3, i, u);
+  }
 }
 
 // Mixed-precision fused optimizer: bf16 gradients in, fp32 master update,
 // bf16 shadow parameters out — one HBM pass covering what autocast training
 // otherwise spends three cast/copy kernel families on.  The atomicAdd's
 // return value folds any concurrently-applied gossip into the shadow.
+template <typename DP>
 __global__ void k_fused_sgd_bf16(float* __restrict__ mom,
                                  const uint16_t* __restrict__ grad,
                                  uint16_t* __restrict__ shadow, float lr,
                                  float momentum, int64_t n,
-                                 float* __restrict__ values, float* d1,
-                                 float* d2, float* d3) {
+                                 float* __restrict__ values,
+                                 typename DP::T* d1, typename DP::T* d2,
+                                 typename DP::T* d3) {
   int64_t gstride = static_cast<int64_t>(gridDim.x) * blockDim.x;
   for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n; i += gstride) {
     float m = momentum * mom[i] + bf16_to_f32(grad[i]);
@@ -411,47 +482,40 @@ __global__ void k_fused_sgd_bf16(float* __restrict__ mom,
     float u = -lr * m;
     float old = atomicAdd(values + i, u);
     shadow[i] = f32_to_bf16(old + u);
-    if (d1) atomicAdd(d1 + i, u);
-    if (d2) atomicAdd(d2 + i, u);
-    if (d3) atomicAdd(d3 + i, u);
-  }
-}
-
-__global__ void k_fused_sgd(float* __restrict__ mom,
-                            const float* __restrict__ grad, float lr,
-                            float momentum, int64_t n, float* d0, float* d1,
-                            float* d2, float* d3) {
-  int64_t gstride = static_cast<int64_t>(gridDim.x) * blockDim.x;
-  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n; i += gstride) {
-    float m = momentum * mom[i] + grad[i];
-    mom[i] = m;
-    float u = -lr * m;
-    if (d0) atomicAdd(d0 + i, u);
-    if (d1) atomicAdd(d1 + i, u);
-    if (d2) atomicAdd(d2 + i, u);
-    if (d3) atomicAdd(d3 + i, u);
+    if (d1) DP::atomic_add(d1, i, u);
+    if (d2) DP::atomic_add(d2, i, u);
+    if (d3) DP::atomic_add(d3, i, u);
   }
 }
 
 // -------------------------------------------------------------- launchers
 
-void hip_reduce_scales(Codec c, const float* delta, const DevTable& tb,
-                       void* reduce_buf, float* scales_out, int stride,
-                       hipStream_t s) {
+void hip_reduce_scales(Codec c, const void* delta, bool delta_bf16,
+                       const DevTable& tb, void* reduce_buf, float* scales_out,
+                       int stride, hipStream_t s) {
   HIP_CHECK(hipMemsetAsync(reduce_buf, 0, tb.T * 8, s));
   // 2048 blocks = 8 per CU (full wave occupancy) while keeping the flush
-  // atomic count at ~8K per launch; each wave register-accumulates across
-  // its grid-stride iterations.
+  // atomic count at ~8K per launch.
   int g = grid_for(tb.pe);
   if (g > 2048) g = 2048;
   if (c == Codec::OneBit) {
-    hipLaunchKernelGGL(k_reduce_sumsq, dim3(g), dim3(BLOCK), 0, s, delta,
-                       tb.offs, tb.poffs, tb.T, tb.pe, stride,
-                       reinterpret_cast<double*>(reduce_buf));
+    if (delta_bf16)
+      hipLaunchKernelGGL((k_reduce_sumsq<DeltaBF16>), dim3(g), dim3(BLOCK), 0, s,
+                         static_cast<const uint16_t*>(delta), tb.offs, tb.poffs,
+                         tb.T, tb.pe, stride, static_cast<double*>(reduce_buf));
+    else
+      hipLaunchKernelGGL((k_reduce_sumsq<DeltaF32>), dim3(g), dim3(BLOCK), 0, s,
+                         static_cast<const float*>(delta), tb.offs, tb.poffs,
+                         tb.T, tb.pe, stride, static_cast<double*>(reduce_buf));
   } else {
-    hipLaunchKernelGGL(k_reduce_absmax, dim3(g), dim3(BLOCK), 0, s, delta,
-                       tb.offs, tb.poffs, tb.T, tb.pe,
-                       reinterpret_cast<uint32_t*>(reduce_buf));
+    if (delta_bf16)
+      hipLaunchKernelGGL((k_reduce_absmax<DeltaBF16>), dim3(g), dim3(BLOCK), 0, s,
+                         static_cast<const uint16_t*>(delta), tb.offs, tb.poffs,
+                         tb.T, tb.pe, static_cast<uint32_t*>(reduce_buf));
+    else
+      hipLaunchKernelGGL((k_reduce_absmax<DeltaF32>), dim3(g), dim3(BLOCK), 0, s,
+                         static_cast<const float*>(delta), tb.offs, tb.poffs,
+                         tb.T, tb.pe, static_cast<uint32_t*>(reduce_buf));
   }
   int gt = (tb.T + BLOCK - 1) / BLOCK;
   hipLaunchKernelGGL(k_finalize_scales, dim3(gt), dim3(BLOCK), 0, s, c,
@@ -467,73 +531,137 @@ void hip_finalize_scales(Codec c, const DevTable& tb, const void* reduce_buf,
   HIP_CHECK(hipGetLastError());
 }
 
-void hip_quantize(Codec c, float* delta, const DevTable& tb,
-                  const float* scales_dev, uint8_t* payload, hipStream_t s,
-                  void* stats_out) {
+template <typename DP>
+static void quantize_dispatch(Codec c, void* delta, const DevTable& tb,
+                              const float* scales_dev, uint8_t* payload,
+                              hipStream_t s, void* stats_out) {
   int g = grid_for(c == Codec::Int4 ? tb.pe / 2 : tb.pe);
   if (stats_out && g > 2048) g = 2048;  // bound the flush-atomic count
+  auto* d = static_cast<typename DP::T*>(delta);
   switch (c) {
     case Codec::OneBit:
-      hipLaunchKernelGGL(k_quant_1bit, dim3(g), dim3(BLOCK), 0, s, delta,
+      hipLaunchKernelGGL((k_quant_1bit<DP>), dim3(g), dim3(BLOCK), 0, s, d,
                          tb.offs, tb.poffs, tb.T, tb.pe, scales_dev,
                          reinterpret_cast<uint64_t*>(payload),
-                         reinterpret_cast<double*>(stats_out));
+                         static_cast<double*>(stats_out));
       break;
     case Codec::Fp8:
-      hipLaunchKernelGGL(k_quant_fp8, dim3(g), dim3(BLOCK), 0, s, delta,
+      hipLaunchKernelGGL((k_quant_fp8<DP>), dim3(g), dim3(BLOCK), 0, s, d,
                          tb.offs, tb.poffs, tb.T, tb.pe, scales_dev, payload,
-                         reinterpret_cast<uint32_t*>(stats_out));
+                         static_cast<uint32_t*>(stats_out));
       break;
     case Codec::Int4:
-      hipLaunchKernelGGL(k_quant_int4, dim3(g), dim3(BLOCK), 0, s, delta,
+      hipLaunchKernelGGL((k_quant_int4<DP>), dim3(g), dim3(BLOCK), 0, s, d,
                          tb.offs, tb.poffs, tb.T, tb.pe, scales_dev, payload,
-                         reinterpret_cast<uint32_t*>(stats_out));
+                         static_cast<uint32_t*>(stats_out));
       break;
   }
+}
+
+void hip_quantize(Codec c, void* delta, bool delta_bf16, const DevTable& tb,
+                  const float* scales_dev, uint8_t* payload, hipStream_t s,
+                  void* stats_out) {
+  if (delta_bf16)
+    quantize_dispatch<DeltaBF16>(c, delta, tb, scales_dev, payload, s, stats_out);
+  else
+    quantize_dispatch<DeltaF32>(c, delta, tb, scales_dev, payload, s, stats_out);
   HIP_CHECK(hipGetLastError());
+}
+
+template <typename DP>
+static void apply_dispatch(Codec c, const uint8_t* payload, const DevTable& tb,
+                           const float* scales_dev, float* values, void* d1,
+                           void* d2, hipStream_t s) {
+  int g = grid_for(tb.pe);
+  auto* t1 = static_cast<typename DP::T*>(d1);
+  auto* t2 = static_cast<typename DP::T*>(d2);
+  switch (c) {
+    case Codec::OneBit:
+      hipLaunchKernelGGL((k_apply<0, DP>), dim3(g), dim3(BLOCK), 0, s, payload,
+                         tb.offs, tb.poffs, tb.T, tb.pe, scales_dev, values, t1, t2);
+      break;
+    case Codec::Fp8:
+      hipLaunchKernelGGL((k_apply<1, DP>), dim3(g), dim3(BLOCK), 0, s, payload,
+                         tb.offs, tb.poffs, tb.T, tb.pe, scales_dev, values, t1, t2);
+      break;
+    case Codec::Int4:
+      hipLaunchKernelGGL((k_apply<2, DP>), dim3(g), dim3(BLOCK), 0, s, payload,
+                         tb.offs, tb.poffs, tb.T, tb.pe, scales_dev, values, t1, t2);
+      break;
+  }
 }
 
 void hip_apply(Codec c, const uint8_t* payload, const DevTable& tb,
-               const float* scales_dev, float* d0, float* d1, float* d2,
-               float* d3, hipStream_t s) {
-  int g = grid_for(tb.pe);
-  switch (c) {
-    case Codec::OneBit:
-      hipLaunchKernelGGL(k_apply<0>, dim3(g), dim3(BLOCK), 0, s, payload,
-                         tb.offs, tb.poffs, tb.T, tb.pe, scales_dev, d0, d1, d2, d3);
-      break;
-    case Codec::Fp8:
-      hipLaunchKernelGGL(k_apply<1>, dim3(g), dim3(BLOCK), 0, s, payload,
-                         tb.offs, tb.poffs, tb.T, tb.pe, scales_dev, d0, d1, d2, d3);
-      break;
-    case Codec::Int4:
-      hipLaunchKernelGGL(k_apply<2>, dim3(g), dim3(BLOCK), 0, s, payload,
-                         tb.offs, tb.poffs, tb.T, tb.pe, scales_dev, d0, d1, d2, d3);
-      break;
-  }
+               const float* scales_dev, float* values, void* d1, void* d2,
+               bool delta_bf16, hipStream_t s) {
+  if (delta_bf16)
+    apply_dispatch<DeltaBF16>(c, payload, tb, scales_dev, values, d1, d2, s);
+  else
+    apply_dispatch<DeltaF32>(c, payload, tb, scales_dev, values, d1, d2, s);
   HIP_CHECK(hipGetLastError());
 }
 
-void hip_add_scatter(const float* src, int64_t n, float alpha, float* d0,
-                     float* d1, float* d2, float* d3, hipStream_t s) {
-  hipLaunchKernelGGL(k_add_scatter, dim3(grid_for(n)), dim3(BLOCK), 0, s, src,
-                     n, alpha, d0, d1, d2, d3);
+void hip_add_scatter(const float* src, int64_t n, float alpha, float* values,
+                     void* d1, void* d2, void* d3, bool delta_bf16,
+                     hipStream_t s) {
+  if (delta_bf16)
+    hipLaunchKernelGGL((k_add_scatter<DeltaBF16>), dim3(grid_for(n)),
+                       dim3(BLOCK), 0, s, src, n, alpha, values,
+                       static_cast<uint16_t*>(d1), static_cast<uint16_t*>(d2),
+                       static_cast<uint16_t*>(d3));
+  else
+    hipLaunchKernelGGL((k_add_scatter<DeltaF32>), dim3(grid_for(n)),
+                       dim3(BLOCK), 0, s, src, n, alpha, values,
+                       static_cast<float*>(d1), static_cast<float*>(d2),
+                       static_cast<float*>(d3));
+  HIP_CHECK(hipGetLastError());
+}
+
+void hip_add_delta_scatter(const void* src_delta, bool delta_bf16, int64_t n,
+                           float* values, void* d1, void* d2, hipStream_t s) {
+  if (delta_bf16)
+    hipLaunchKernelGGL((k_add_delta_scatter<DeltaBF16>), dim3(grid_for(n)),
+                       dim3(BLOCK), 0, s,
+                       static_cast<const uint16_t*>(src_delta), n, values,
+                       static_cast<uint16_t*>(d1), static_cast<uint16_t*>(d2));
+  else
+    hipLaunchKernelGGL((k_add_delta_scatter<DeltaF32>), dim3(grid_for(n)),
+                       dim3(BLOCK), 0, s, static_cast<const float*>(src_delta),
+                       n, values, static_cast<float*>(d1),
+                       static_cast<float*>(d2));
   HIP_CHECK(hipGetLastError());
 }
 
 void hip_fused_sgd(float* mom, const float* grad, float lr, float momentum,
-                   int64_t n, float* d0, float* d1, float* d2, float* d3,
-                   hipStream_t s) {
-  hipLaunchKernelGGL(k_fused_sgd, dim3(grid_for(n)), dim3(BLOCK), 0, s, mom,
-                     grad, lr, momentum, n, d0, d1, d2, d3);
+                   int64_t n, float* values, void* d1, void* d2, void* d3,
+                   bool delta_bf16, hipStream_t s) {
+  if (delta_bf16)
+    hipLaunchKernelGGL((k_fused_sgd<DeltaBF16>), dim3(grid_for(n)), dim3(BLOCK),
+                       0, s, mom, grad, lr, momentum, n, values,
+                       static_cast<uint16_t*>(d1), static_cast<uint16_t*>(d2),
+                       static_cast<uint16_t*>(d3));
+  else
+    hipLaunchKernelGGL((k_fused_sgd<DeltaF32>), dim3(grid_for(n)), dim3(BLOCK),
+                       0, s, mom, grad, lr, momentum, n, values,
+                       static_cast<float*>(d1), static_cast<float*>(d2),
+                       static_cast<float*>(d3));
   HIP_CHECK(hipGetLastError());
 }
 
 void hip_fused_sgd_bf16(float* mom, const uint16_t* grad, uint16_t* shadow,
                         float lr, float momentum, int64_t n, float* values,
-                        float* d1, float* d2, float* d3, hipStream_t s) {
-  hipLaunchKernelGGL(k_fused_sgd_bf16, dim3(grid_for(n)), dim3(BLOCK), 0, s,
-                     mom, grad, shadow, lr, momentum, n, values, d1, d2, d3);
+                        void* d1, void* d2, void* d3, bool delta_bf16,
+                        hipStream_t s) {
+  if (delta_bf16)
+    hipLaunchKernelGGL((k_fused_sgd_bf16<DeltaBF16>), dim3(grid_for(n)),
+                       dim3(BLOCK), 0, s, mom, grad, shadow, lr, momentum, n,
+                       values, static_cast<uint16_t*>(d1),
+                       static_cast<uint16_t*>(d2), static_cast<uint16_t*>(d3));
+  else
+    hipLaunchKernelGGL((k_fused_sgd_bf16<DeltaF32>), dim3(grid_for(n)),
+                       dim3(BLOCK), 0, s, mom, grad, shadow, lr, momentum, n,
+                       values, static_cast<float*>(d1), static_cast<float*>(d2),
+                       static_cast<float*>(d3));
   HIP_CHECK(hipGetLastError());
 }
 

@@ -42,7 +42,8 @@ class _SharedBase:
                  provision_up: bool = True, explicit_parent: str = "",
                  listen_port: int = 0, join_timeout_s: float = 60.0,
                  rms_sample_stride: int = 1, lagged_scale: bool = False,
-                 use_graphs: bool = True):
+                 delta_dtype: torch.dtype = torch.float32,
+                 use_graphs: bool = False):
         if codec not in CODECS:
             raise ValueError(f"codec must be one of {list(CODECS)}")
         self.device = torch.device(device)
@@ -64,6 +65,12 @@ class _SharedBase:
         cfg.join_timeout_s = join_timeout_s
         cfg.rms_sample_stride = rms_sample_stride
         cfg.lagged_scale = bool(lagged_scale)
+        if delta_dtype not in (torch.float32, torch.bfloat16):
+            raise ValueError("delta_dtype must be float32 or bfloat16")
+        if delta_dtype == torch.bfloat16 and not self._gpu:
+            raise ValueError("bf16 residual deltas need a GPU engine")
+        cfg.delta_bf16 = delta_dtype == torch.bfloat16
+        self.delta_dtype = delta_dtype
         cfg.use_graphs = bool(use_graphs)
         self.codec = codec
         self.n = int(sum(sizes))
@@ -83,7 +90,8 @@ class _SharedBase:
         for li in (0, 1, 2):
             if li not in links:
                 continue
-            delta = torch.zeros(self.n, dtype=torch.float32, device=self.device)
+            delta = torch.zeros(self.n, dtype=self.delta_dtype,
+                                device=self.device)
             if self._gpu:
                 send_buf = torch.empty(msg, dtype=torch.uint8, device=self.device)
                 recv_buf = torch.empty(msg, dtype=torch.uint8, device=self.device)

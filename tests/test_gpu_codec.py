@@ -38,7 +38,7 @@ def test_encode_matches_oracle(codec, n):
 
 
 @pytest.mark.parametrize("codec", [0, 1, 2])
-@pytest.mark.parametrize("ndst", [1, 2, 4])
+@pytest.mark.parametrize("ndst", [1, 2, 3])
 def test_apply_matches_oracle(codec, ndst):
     n = 1 << 18
     torch.manual_seed(codec * 7 + ndst)
@@ -167,6 +167,40 @@ def test_lagged_scale_stats(codec):
     dc.reduce_scales(d.data_ptr(), fresh.data_ptr(), 1, s)
     torch.cuda.synchronize()
     assert lagged.item() == fresh.item(), (lagged.item(), fresh.item())
+
+
+@pytest.mark.parametrize("codec", [0, 1, 2])
+def test_bf16_delta_roundtrip(codec):
+    """bf16 residual storage: the EF debit quanta are exactly representable
+    in bf16 (pow2 scales x small-mantissa q), so residual' == bf16(residual)
+    - sent exactly when the pre-quantize residual is itself bf16."""
+    n = 1 << 16
+    torch.manual_seed(31 + codec)
+    d32 = (torch.randn(n) * 2)
+    d = d32.to(torch.bfloat16).cuda()
+    dc = _core.DevCodec(codec, [n], 0, delta_bf16=True)
+    scales = torch.zeros(1, dtype=torch.float32, device="cuda")
+    payload = torch.zeros(_core.payload_bytes(codec, n), dtype=torch.uint8,
+                          device="cuda")
+    dc.reduce_scales(d.data_ptr(), scales.data_ptr(), 1, stream())
+    dc.quantize(d.data_ptr(), scales.data_ptr(), payload.data_ptr(), stream())
+    torch.cuda.synchronize()
+    s = scales.item()
+    assert s > 0
+    dec = oc.decode(codec, payload.cpu().numpy().tobytes(), s, n)
+    # conservation in bf16: residual_after == bf16(fl32(bf16(d) - sent))
+    expect = (d.cpu().float() - dec).to(torch.bfloat16)
+    assert torch.equal(d.cpu(), expect), \
+        (d.cpu().float() - expect.float()).abs().max()
+    # and apply into a bf16 destination accumulates
+    dst = torch.zeros(n, dtype=torch.bfloat16, device="cuda")
+    vals = torch.zeros(n, dtype=torch.float32, device="cuda")
+    dc.apply(payload.data_ptr(), scales.data_ptr(),
+             [vals.data_ptr(), dst.data_ptr()], stream())
+    torch.cuda.synchronize()
+    torch.testing.assert_close(vals.cpu(), dec, rtol=0, atol=0)
+    torch.testing.assert_close(dst.cpu().float(), dec.to(torch.bfloat16).float(),
+                               rtol=0, atol=0)
 
 
 def test_quantize_keepalive_zero_scale():

@@ -33,12 +33,13 @@ def wait_until(fn, timeout=30.0, interval=0.05):
     return False
 
 
-def _gpu_child(port, q, codec, lagged=False):
+def _gpu_child(port, q, codec, lagged=False, bf16d=False):
     try:
         torch.cuda.set_device(0)
         seed = torch.zeros(1 << 20, device="cuda")
         h = st.create_or_fetch("127.0.0.1", port, seed, codec=codec,
-                               lagged_scale=lagged)
+                               lagged_scale=lagged,
+                               delta_dtype=torch.bfloat16 if bf16d else torch.float32)
         target = torch.full((1 << 20,), 3.0, device="cuda")
         out = torch.zeros_like(seed)
 
@@ -58,17 +59,20 @@ def _gpu_child(port, q, codec, lagged=False):
         q.put(("fail", repr(e)))
 
 
-@pytest.mark.parametrize("codec,lagged", [("1bit", False), ("fp8", False),
-                                          ("1bit", True), ("int4", True)])
-def test_two_process_one_gpu_tcp(codec, lagged):
+@pytest.mark.parametrize("codec,lagged,bf16d",
+                         [("1bit", False, False), ("fp8", False, False),
+                          ("1bit", True, False), ("int4", True, False),
+                          ("1bit", True, True), ("fp8", False, True)])
+def test_two_process_one_gpu_tcp(codec, lagged, bf16d):
     port = free_port()
     ctx = mp.get_context("spawn")
     torch.cuda.set_device(0)
     master = st.create_or_fetch("127.0.0.1", port,
                                 torch.full((1 << 20,), 3.0, device="cuda"),
-                                codec=codec, lagged_scale=lagged)
+                                codec=codec, lagged_scale=lagged,
+                                delta_dtype=torch.bfloat16 if bf16d else torch.float32)
     q = ctx.Queue()
-    p = ctx.Process(target=_gpu_child, args=(port, q, codec, lagged))
+    p = ctx.Process(target=_gpu_child, args=(port, q, codec, lagged, bf16d))
     p.start()
     try:
         status, msg = q.get(timeout=120)
