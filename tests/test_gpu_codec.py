@@ -178,6 +178,7 @@ def test_bf16_delta_roundtrip(codec):
     torch.manual_seed(31 + codec)
     d32 = (torch.randn(n) * 2)
     d = d32.to(torch.bfloat16).cuda()
+    d_orig = d.cpu().clone()  # quantize debits d in place
     dc = _core.DevCodec(codec, [n], 0, delta_bf16=True)
     scales = torch.zeros(1, dtype=torch.float32, device="cuda")
     payload = torch.zeros(_core.payload_bytes(codec, n), dtype=torch.uint8,
@@ -188,8 +189,8 @@ def test_bf16_delta_roundtrip(codec):
     s = scales.item()
     assert s > 0
     dec = oc.decode(codec, payload.cpu().numpy().tobytes(), s, n)
-    # conservation in bf16: residual_after == bf16(fl32(bf16(d) - sent))
-    expect = (d.cpu().float() - dec).to(torch.bfloat16)
+    # conservation in bf16: residual_after == bf16(fl32(bf16(d)) - sent)
+    expect = (d_orig.float() - dec).to(torch.bfloat16)
     assert torch.equal(d.cpu(), expect), \
         (d.cpu().float() - expect.float()).abs().max()
     # and apply into a bf16 destination accumulates
