@@ -13,23 +13,11 @@ import pytest
 import torch
 
 import sharedtensor_amd as st
+from sharedtensor_amd.utils import free_port, wait_until
 
 
-def free_port():
-    s = socket.socket()
-    s.bind(("127.0.0.1", 0))
-    p = s.getsockname()[1]
-    s.close()
-    return p
 
 
-def wait_until(fn, timeout=60.0, interval=0.05):
-    t0 = time.time()
-    while time.time() - t0 < timeout:
-        if fn():
-            return True
-        time.sleep(interval)
-    return False
 
 
 N = 2048

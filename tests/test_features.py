@@ -9,14 +9,9 @@ import pytest
 import torch
 
 import sharedtensor_amd as st
+from sharedtensor_amd.utils import free_port, wait_until
 
 
-def free_port():
-    s = socket.socket()
-    s.bind(("127.0.0.1", 0))
-    p = s.getsockname()[1]
-    s.close()
-    return p
 
 
 def _bw_child(port, q, bw_limit):
