@@ -81,6 +81,13 @@ def run_train(args, rank, world, device):
     cfg = GPT2Config.tiny() if args.model == "tiny" else GPT2Config.small()
     if args.seq:
         cfg.block_size = min(cfg.block_size, args.seq) if args.model == "tiny" else args.seq
+    fa = os.environ.get("SHTENS_FA", "")
+    if fa and device.type == "cuda":
+        try:
+            torch.backends.cuda.preferred_rocm_fa_library(fa)
+            log(f"rocm flash-attention backend: {fa}")
+        except Exception as e:
+            log(f"fa backend {fa} unavailable: {e}")
     torch.manual_seed(1234)  # same random init on every rank
     model = GPT2(cfg).to(device)
     log(f"model {args.model}: {model.num_params()/1e6:.1f}M params, device {device}")
