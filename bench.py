@@ -336,7 +336,7 @@ def main():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=20)
     ap.add_argument("--warmup", type=int, default=5)
-    ap.add_argument("--batch", type=int, default=32)
+    ap.add_argument("--batch", type=int, default=64)
     ap.add_argument("--seq", type=int, default=1024)
     ap.add_argument("--lr", type=float, default=0.01)
     ap.add_argument("--codec", choices=["1bit", "fp8", "int4"], default="1bit")

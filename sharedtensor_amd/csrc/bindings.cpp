@@ -63,9 +63,9 @@ struct DevCodec {
     if (hipMalloc(&tb.offs, sizeof(int64_t) * (T + 1) * 2) != hipSuccess)
       throw std::runtime_error("hipMalloc failed");
     tb.poffs = tb.offs + (T + 1);
-    hipMemcpy(tb.offs, offs.data(), sizeof(int64_t) * (T + 1),
+    (void)hipMemcpy(tb.offs, offs.data(), sizeof(int64_t) * (T + 1),
               hipMemcpyHostToDevice);
-    hipMemcpy(tb.poffs, poffs.data(), sizeof(int64_t) * (T + 1),
+    (void)hipMemcpy(tb.poffs, poffs.data(), sizeof(int64_t) * (T + 1),
               hipMemcpyHostToDevice);
     tb.T = T;
     tb.n = offs[T];
@@ -74,8 +74,8 @@ struct DevCodec {
       throw std::runtime_error("hipMalloc failed");
   }
   ~DevCodec() {
-    if (tb.offs) hipFree(tb.offs);
-    if (reduce) hipFree(reduce);
+    if (tb.offs) (void)hipFree(tb.offs);
+    if (reduce) (void)hipFree(reduce);
   }
   void reduce_scales(uintptr_t delta, uintptr_t scales_dev, int stride,
                      uintptr_t stream) {

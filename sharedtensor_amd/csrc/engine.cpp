@@ -248,11 +248,11 @@ void Engine::free_gpu() {
   if (!gpu()) return;
   for (auto& lk : links_) {
     destroy_link_graphs(lk);
-    if (lk.s_send) hipStreamDestroy(lk.s_send), lk.s_send = nullptr;
-    if (lk.s_recv) hipStreamDestroy(lk.s_recv), lk.s_recv = nullptr;
-    if (lk.reduce_buf) hipFree(lk.reduce_buf), lk.reduce_buf = nullptr;
+    if (lk.s_send) (void)hipStreamDestroy(lk.s_send), lk.s_send = nullptr;
+    if (lk.s_recv) (void)hipStreamDestroy(lk.s_recv), lk.s_recv = nullptr;
+    if (lk.reduce_buf) (void)hipFree(lk.reduce_buf), lk.reduce_buf = nullptr;
   }
-  if (dtb_.offs) hipFree(dtb_.offs), dtb_.offs = nullptr;
+  if (dtb_.offs) (void)hipFree(dtb_.offs), dtb_.offs = nullptr;
 }
 
 // ------------------------------------------------------------------ start
@@ -996,7 +996,7 @@ bool Engine::send_packet(Link& lk, const float* scales_host) {
 }
 
 void Engine::send_loop(Link& lk) {
-  if (gpu()) hipSetDevice(cfg_.device);
+  if (gpu()) (void)hipSetDevice(cfg_.device);
   std::vector<float> scales(T_);
   auto last_send = Clock::now();
   auto next_allowed = Clock::now();
@@ -1129,7 +1129,7 @@ void Engine::apply_packet(Link& lk, const float* scales_host) {
 }
 
 void Engine::recv_loop(Link& lk) {
-  if (gpu()) hipSetDevice(cfg_.device);
+  if (gpu()) (void)hipSetDevice(cfg_.device);
   std::vector<float> scales(T_);
   while (!closing_ && lk.state.load() == L_ACTIVE) {
     if (lk.rccl) {

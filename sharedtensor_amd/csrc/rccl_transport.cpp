@@ -117,7 +117,7 @@ static bool wait_stream(RcclLink* l, ncclComm_t c, hipStream_t s,
     if (abort.load() || l->aborted.load()) {
       ncclCommAbort(c);
       l->aborted.store(true);
-      hipStreamSynchronize(s);
+      (void)hipStreamSynchronize(s);
       return false;
     }
     std::this_thread::sleep_for(std::chrono::microseconds(20));
@@ -161,23 +161,23 @@ void rccl_self_test(int device) {
   float* buf = nullptr;
   if (hipMalloc(&buf, 1024 * 4) != hipSuccess)
     throw std::runtime_error("hipMalloc failed");
-  hipMemset(buf, 0, 1024 * 4);
+  (void)hipMemset(buf, 0, 1024 * 4);
   hipStream_t s;
-  hipStreamCreate(&s);
+  (void)hipStreamCreate(&s);
   ncclResult_t r = ncclAllReduce(buf, buf, 1024, ncclFloat, ncclSum, c, s);
   if (r != ncclSuccess && r != ncclInProgress) {
-    hipFree(buf);
+    (void)hipFree(buf);
     throw std::runtime_error(std::string("self-test allreduce: ") +
                              ncclGetErrorString(r));
   }
   wait_comm(c, 30.0, "self-test allreduce enqueue");
   if (hipStreamSynchronize(s) != hipSuccess) {
-    hipFree(buf);
+    (void)hipFree(buf);
     throw std::runtime_error("self-test stream sync failed");
   }
-  hipStreamDestroy(s);
-  hipFree(buf);
-  ncclCommDestroy(c);
+  (void)hipStreamDestroy(s);
+  (void)hipFree(buf);
+  (void)ncclCommDestroy(c);
 }
 
 void rccl_destroy(RcclLink* l) {
