@@ -331,8 +331,10 @@ void Engine::join_tree() {
     Hello h{};
     h.magic = MAGIC;
     h.version = PROTO_VERSION;
-    h.flags = static_cast<uint16_t>((gpu() ? HELLO_HAS_GPU : 0) |
-                                    (gpu() && cfg_.use_rccl ? HELLO_WANT_RCCL : 0));
+    h.flags = static_cast<uint16_t>(
+        (gpu() ? HELLO_HAS_GPU : 0) |
+        (gpu() && cfg_.use_rccl && rccl_failures_.load() < 2 ? HELLO_WANT_RCCL
+                                                             : 0));
     h.n = static_cast<uint64_t>(n_);
     h.ntensors = static_cast<uint32_t>(T_);
     h.codec = static_cast<uint32_t>(cfg_.codec);
@@ -365,8 +367,30 @@ void Engine::join_tree() {
       throw std::runtime_error("protocol error during join (reply byte " +
                                std::to_string(int(reply)) + ")");
     }
-    handshake_as_child(fd);
-    return;
+    try {
+      handshake_as_child(fd);
+      return;
+    } catch (const std::exception& e) {
+      // transient handshake failure (e.g. RCCL init): close, back off and
+      // retry the walk — after 2 RCCL failures the hello stops asking
+      ::close(fd);
+      if (listen_fd_ >= 0) {  // listener was bound in the failed attempt
+        ::close(listen_fd_);
+        listen_fd_ = -1;
+      }
+      Link& up0 = links_[LK_UP];
+      up0.fd = -1;
+      // a partial snapshot may have been applied; reset to the pristine
+      // pre-join state (all-zero) so the retry cannot double-count
+      zero_buf(values_, n_);
+      for (auto& l0 : links_)
+        if (l0.provisioned) zero_delta(l0.delta);
+      set_error(std::string("join handshake failed, retrying: ") + e.what());
+      std::this_thread::sleep_for(std::chrono::milliseconds(300));
+      target = explicit_mode ? target : root_addr_;
+      hops_since_root = 0;
+      continue;
+    }
   }
 }
 
@@ -413,7 +437,12 @@ void Engine::handshake_as_child(int fd, bool rejoin) {
     uint8_t ids[2 * RCCL_ID_BYTES];
     if (!io_read(fd, ids, sizeof(ids)))
       throw std::runtime_error("failed to read RCCL ids from parent");
-    rccl_upgrade(up, ids, /*is_parent=*/false);
+    try {
+      rccl_upgrade(up, ids, /*is_parent=*/false);
+    } catch (...) {
+      rccl_failures_++;  // retried joins fall back to TCP after 2 failures
+      throw;
+    }
   }
   if (rejoin && (ah.flags & ACC_SNAPSHOT)) {
     // reconciliation on rejoin: V := S + tmp, where S is the new parent's
@@ -596,8 +625,10 @@ void Engine::reconnect_loop() try {
     Hello h{};
     h.magic = MAGIC;
     h.version = PROTO_VERSION;
-    h.flags = static_cast<uint16_t>((gpu() ? HELLO_HAS_GPU : 0) |
-                                    (gpu() && cfg_.use_rccl ? HELLO_WANT_RCCL : 0));
+    h.flags = static_cast<uint16_t>(
+        (gpu() ? HELLO_HAS_GPU : 0) |
+        (gpu() && cfg_.use_rccl && rccl_failures_.load() < 2 ? HELLO_WANT_RCCL
+                                                             : 0));
     h.n = static_cast<uint64_t>(n_);
     h.ntensors = static_cast<uint32_t>(T_);
     h.codec = static_cast<uint32_t>(cfg_.codec);
@@ -802,7 +833,11 @@ void Engine::accept_child(int fd, const Hello& h, const sockaddr_in& peer,
         throw std::runtime_error("failed to send RCCL ids");
       rccl_upgrade(lk, ids, /*is_parent=*/true);
     } catch (const std::exception& e) {
-      set_error(std::string("rccl upgrade failed: ") + e.what());
+      // after repeated failures both sides stop negotiating RCCL and the
+      // joiner's retry lands on the plain TCP data plane
+      rccl_failures_++;
+      set_error(std::string("rccl upgrade failed (") +
+                std::to_string(rccl_failures_.load()) + "): " + e.what());
       ::close(fd);
       lk.fd = -1;
       return;
@@ -829,12 +864,14 @@ void Engine::spawn_link_threads(Link& lk) {
 }
 
 bool Engine::rccl_wanted(const Hello& h) const {
-  return gpu() && cfg_.use_rccl && (h.flags & HELLO_WANT_RCCL) &&
-         h.hostid == hostid_ && h.device >= 0 && h.device != cfg_.device;
+  return gpu() && cfg_.use_rccl && rccl_failures_.load() < 2 &&
+         (h.flags & HELLO_WANT_RCCL) && h.hostid == hostid_ && h.device >= 0 &&
+         h.device != cfg_.device;
 }
 
 void Engine::rccl_upgrade(Link& lk, const uint8_t* ids, bool is_parent) {
-  lk.rccl_link = rccl_link_create(cfg_.device, ids, is_parent, cfg_.join_timeout_s);
+  double to = cfg_.join_timeout_s < 20.0 ? cfg_.join_timeout_s : 20.0;
+  lk.rccl_link = rccl_link_create(cfg_.device, ids, is_parent, to);
   lk.rccl = true;
 }
 

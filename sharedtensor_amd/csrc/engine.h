@@ -109,6 +109,7 @@ class Engine {
   std::thread reconnect_thread_;
   std::atomic<bool> reconnecting_{false};
   std::atomic<uint64_t> reconnects_{0};
+  std::atomic<int> rccl_failures_{0};  // >=2: stop offering/requesting RCCL
   std::atomic<bool> closing_{false};
   bool started_ = false;
   bool is_master_ = false;
