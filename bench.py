@@ -148,7 +148,8 @@ def run_train(args, rank, world, device):
     tokens = float(B * T * args.steps * world)
     value = tokens / dt
     result = {
-        "metric": "async-DP tokens/sec GPT-2-small",
+        "metric": ("async-DP tokens/sec GPT-2-small" if args.model in ("small", "tiny")
+                   else f"async-DP tokens/sec {args.model}"),
         "value": round(value, 1),
         "unit": "tokens/s",
         "n_gpus": world,
