@@ -850,6 +850,11 @@ void Engine::accept_child(int fd, const Hello& h, const sockaddr_in& peer,
       set_error(std::string("snapshot send failed: ") + e.what());
       ::close(fd);
       lk.fd = -1;
+      if (lk.rccl_link) {  // don't leak the freshly-made comm pair
+        rccl_destroy(static_cast<RcclLink*>(lk.rccl_link));
+        lk.rccl_link = nullptr;
+        lk.rccl = false;
+      }
       return;
     }
   }
