@@ -103,6 +103,7 @@ def run_train(args, rank, world, device):
         lagged_scale=(not args.exact_scale) and device.type == "cuda",
         use_graphs=args.graphs,
         delta_dtype=torch.bfloat16 if args.bf16_deltas else torch.float32,
+        sync_interval_s=args.sync_interval,
         snapshot_join=True)
 
     B, T = args.batch, cfg.block_size
@@ -345,6 +346,10 @@ def main():
     ap.add_argument("--mode", choices=["train", "paramsync", "table"], default="train")
     ap.add_argument("--numel", type=int, default=268_435_456)  # 1 GB fp32
     ap.add_argument("--interval", type=float, default=0.01)
+    ap.add_argument("--sync-interval", type=float, default=0.01,
+                    help="min seconds between a link's sync rounds (0 = "
+                         "free-run like the reference; pacing keeps the codec "
+                         "kernels from starving training compute of HBM)")
     ap.add_argument("--no-rccl", action="store_true")
     ap.add_argument("--fp32-params", action="store_true",
                     help="compute on fp32 replica views (default: bf16 shadow)")

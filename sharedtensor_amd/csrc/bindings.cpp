@@ -154,6 +154,7 @@ PYBIND11_MODULE(_core, m) {
       .def_readwrite("reconnect", &Config::reconnect)
       .def_readwrite("keepalive_s", &Config::keepalive_s)
       .def_readwrite("bw_limit", &Config::bw_limit)
+      .def_readwrite("min_round_interval_s", &Config::min_round_interval_s)
       .def_readwrite("expected_children", &Config::expected_children)
       .def_readwrite("sizes", &Config::sizes)
       .def_readwrite("explicit_parent", &Config::explicit_parent)

@@ -51,6 +51,11 @@ struct Config {
   double keepalive_s = 1.0;  // idle ping interval (reference: 1s, :161-164)
   double bw_limit = 0.0;     // bytes/sec cap per link, 0 = unlimited
                              // (reference TODO, README.md:31)
+  double min_round_interval_s = 0.0;  // pace each link's send loop: the
+                             // reference free-runs ("fills all bandwidth"),
+                             // but on GPU the codec kernels share HBM with
+                             // training compute — pacing trades (already
+                             // tiny) staleness for compute bandwidth
   int expected_children = 2; // how many child links to provision buffers for
   std::vector<int64_t> sizes;  // per-tensor element counts (table mode;
                                // size 1 = classic flat tensor)

@@ -1076,13 +1076,17 @@ void Engine::send_loop(Link& lk) {
       lk.dirty = false;
       continue;
     }
-    // bandwidth cap (reference TODO, README.md:31): token-bucket pacing
-    if (cfg_.bw_limit > 0) {
+    // pacing: bandwidth cap (reference TODO, README.md:31) and/or a
+    // minimum per-round interval (HBM-bandwidth courtesy towards compute)
+    if (cfg_.bw_limit > 0 || cfg_.min_round_interval_s > 0) {
       auto now = Clock::now();
       if (now < next_allowed)
         std::this_thread::sleep_for(next_allowed - now);
+      double gap = cfg_.min_round_interval_s;
+      if (cfg_.bw_limit > 0)
+        gap = std::max(gap, (8.0 + SA_ + P_) / cfg_.bw_limit);
       next_allowed = Clock::now() + std::chrono::duration_cast<Clock::duration>(
-          std::chrono::duration<double>((8.0 + SA_ + P_) / cfg_.bw_limit));
+          std::chrono::duration<double>(gap));
     }
     try {
       if (gpu()) {

@@ -38,7 +38,8 @@ class _SharedBase:
                  device: torch.device, codec: str = "1bit", *,
                  snapshot_join: bool = True, use_rccl: bool = True,
                  reconnect: bool = False, keepalive_s: float = 1.0,
-                 bw_limit: float = 0.0, expected_children: int = 2,
+                 bw_limit: float = 0.0, sync_interval_s: float = 0.0,
+                 expected_children: int = 2,
                  provision_up: bool = True, explicit_parent: str = "",
                  listen_port: int = 0, join_timeout_s: float = 60.0,
                  rms_sample_stride: int = 1, lagged_scale: bool = False,
@@ -58,6 +59,7 @@ class _SharedBase:
         cfg.reconnect = reconnect
         cfg.keepalive_s = keepalive_s
         cfg.bw_limit = float(bw_limit)
+        cfg.min_round_interval_s = float(sync_interval_s)
         cfg.expected_children = expected_children
         cfg.sizes = [int(s) for s in sizes]
         cfg.explicit_parent = explicit_parent
