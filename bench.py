@@ -346,7 +346,7 @@ def main():
     ap.add_argument("--mode", choices=["train", "paramsync", "table"], default="train")
     ap.add_argument("--numel", type=int, default=268_435_456)  # 1 GB fp32
     ap.add_argument("--interval", type=float, default=0.01)
-    ap.add_argument("--sync-interval", type=float, default=0.01,
+    ap.add_argument("--sync-interval", type=float, default=0.05,
                     help="min seconds between a link's sync rounds (0 = "
                          "free-run like the reference; pacing keeps the codec "
                          "kernels from starving training compute of HBM)")
