@@ -26,6 +26,7 @@ SOURCES = [
     "engine.cpp",
     "rccl_transport.cpp",
     "hip_kernels.hip",
+    "ln_kernels.hip",
     "bindings.cpp",
 ]
 

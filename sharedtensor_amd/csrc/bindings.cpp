@@ -223,6 +223,31 @@ PYBIND11_MODULE(_core, m) {
   m.def("rccl_self_test", &rccl_self_test,
         py::call_guard<py::gil_scoped_release>());
 
+  // fused bf16 LayerNorm (pointers are bf16 unless named mean/rstd/dgamma/
+  // dbeta, which are fp32)
+  m.def("ln_fwd", [](uintptr_t x, uintptr_t w, uintptr_t b, uintptr_t y,
+                     uintptr_t mean, uintptr_t rstd, int64_t R, int C,
+                     double eps, uintptr_t stream) {
+    hip_ln_fwd(reinterpret_cast<const void*>(x),
+               reinterpret_cast<const void*>(w),
+               reinterpret_cast<const void*>(b), reinterpret_cast<void*>(y),
+               reinterpret_cast<float*>(mean), reinterpret_cast<float*>(rstd),
+               R, C, static_cast<float>(eps),
+               reinterpret_cast<hipStream_t>(stream));
+  });
+  m.def("ln_bwd", [](uintptr_t dy, uintptr_t x, uintptr_t w, uintptr_t mean,
+                     uintptr_t rstd, uintptr_t dx, uintptr_t dgamma,
+                     uintptr_t dbeta, int64_t R, int C, uintptr_t stream) {
+    hip_ln_bwd(reinterpret_cast<const void*>(dy),
+               reinterpret_cast<const void*>(x),
+               reinterpret_cast<const void*>(w),
+               reinterpret_cast<const float*>(mean),
+               reinterpret_cast<const float*>(rstd),
+               reinterpret_cast<void*>(dx), reinterpret_cast<float*>(dgamma),
+               reinterpret_cast<float*>(dbeta), R, C,
+               reinterpret_cast<hipStream_t>(stream));
+  });
+
   m.def("msg_bytes", &Engine::msg_bytes);
   m.def("scales_area", &Engine::scales_area);
   m.def("payload_bytes",

@@ -83,4 +83,13 @@ void hip_fused_sgd_bf16(float* mom, const uint16_t* grad, uint16_t* shadow,
                         void* d1, void* d2, void* d3, bool delta_bf16,
                         hipStream_t s);
 
+// Fused bf16 LayerNorm for the training path (ln_kernels.hip): fwd saves
+// fp32 mean/rstd; bwd = dx pass + register-accumulated dgamma/dbeta pass.
+void hip_ln_fwd(const void* x, const void* w, const void* b, void* y,
+                float* mean, float* rstd, int64_t R, int C, float eps,
+                hipStream_t s);
+void hip_ln_bwd(const void* dy, const void* x, const void* w,
+                const float* mean, const float* rstd, void* dx, float* dgamma,
+                float* dbeta, int64_t R, int C, hipStream_t s);
+
 }  // namespace shamd

@@ -51,6 +51,10 @@ class FusedLayerNorm(nn.LayerNorm):
     def forward(self, x):
         if (x.is_cuda and x.dtype == torch.bfloat16
                 and self.weight.dtype == torch.bfloat16):
+            from ..ops import fused_ln
+            if fused_ln.can_use(x, self.weight):
+                return fused_ln.fused_layer_norm(x, self.weight, self.bias,
+                                                 self.eps)
             with torch.autocast("cuda", enabled=False):
                 return F.layer_norm(x, self.normalized_shape, self.weight,
                                     self.bias, self.eps)
