@@ -61,10 +61,6 @@ def test_training_equivalence_small_model():
         torch.manual_seed(7)
         cfg = GPT2Config.tiny()
         m = GPT2(cfg).cuda().to(torch.bfloat16)
-        if tag == "torch":
-            # disable the fused path by masquerading the dtype check
-            for blk in m.blocks:
-                blk.ln1.float().to(torch.bfloat16)
         opt = torch.optim.SGD(m.parameters(), lr=0.05)
         x = torch.randint(0, cfg.vocab_size, (2, 33), device="cuda")
         ls = []
