@@ -74,7 +74,7 @@ def test_training_equivalence_small_model():
                 _, loss = m(x[:, :-1], x[:, 1:])
                 loss.float().backward()
                 opt.step()
-                ls.append(float(loss))
+                ls.append(float(loss.detach()))
         finally:
             fl.can_use = orig
         losses[tag] = ls
