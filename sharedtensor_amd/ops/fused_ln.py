@@ -39,8 +39,8 @@ class _FusedLayerNormFn(torch.autograd.Function):
         C = x.shape[-1]
         R = x.numel() // C
         dx = torch.empty_like(x)
-        dgamma = torch.zeros(C, dtype=torch.float32, device=x.device)
-        dbeta = torch.zeros(C, dtype=torch.float32, device=x.device)
+        dwdb = torch.zeros(2 * C, dtype=torch.float32, device=x.device)
+        dgamma, dbeta = dwdb[:C], dwdb[C:]
         s = torch.cuda.current_stream(x.device).cuda_stream
         _core.ln_bwd(dy.data_ptr(), x.data_ptr(), w.data_ptr(),
                      mean.data_ptr(), rstd.data_ptr(), dx.data_ptr(),
