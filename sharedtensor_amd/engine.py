@@ -147,18 +147,21 @@ class _SharedBase:
         return 0
 
     def _add_flat(self, flat: torch.Tensor):
-        assert flat.dtype == torch.float32 and flat.is_contiguous()
+        if flat.dtype != torch.float32 or not flat.is_contiguous():
+            raise TypeError("expected a contiguous float32 tensor")
         self._eng.add_from(flat.data_ptr(), flat.numel(), self._stream())
 
     def _copy_flat(self, flat: torch.Tensor):
-        assert flat.dtype == torch.float32 and flat.is_contiguous()
+        if flat.dtype != torch.float32 or not flat.is_contiguous():
+            raise TypeError("expected a contiguous float32 tensor")
         self._eng.copy_to(flat.data_ptr(), flat.numel(), self._stream())
 
     def fused_sgd_step(self, momentum_buf: torch.Tensor, grad: torch.Tensor,
                        lr: float, momentum: float = 0.9):
         """m = mu*m + g; u = -lr*m; {replica, link deltas} += u — one fused
         HBM pass (HIP kernel k_fused_sgd) instead of optimizer + addFromTensor."""
-        assert momentum_buf.numel() == self.n and grad.numel() == self.n
+        if momentum_buf.numel() != self.n or grad.numel() != self.n:
+            raise ValueError("size mismatch")
         self._eng.fused_sgd(momentum_buf.data_ptr(), grad.data_ptr(),
                             float(lr), float(momentum), self._stream())
 
@@ -168,8 +171,11 @@ class _SharedBase:
         """Mixed-precision fused step: bf16 grads in, fp32 master updated,
         bf16 shadow params refreshed, link deltas staged — one HBM pass
         (HIP kernel k_fused_sgd_bf16)."""
-        assert grad_bf16.dtype == torch.bfloat16 and shadow_bf16.dtype == torch.bfloat16
-        assert momentum_buf.numel() == grad_bf16.numel() == shadow_bf16.numel() == self.n
+        if grad_bf16.dtype != torch.bfloat16 or shadow_bf16.dtype != torch.bfloat16:
+            raise TypeError("grad/shadow must be bfloat16")
+        if not (momentum_buf.numel() == grad_bf16.numel()
+                == shadow_bf16.numel() == self.n):
+            raise ValueError("size mismatch")
         self._eng.fused_sgd_bf16(momentum_buf.data_ptr(), grad_bf16.data_ptr(),
                                  shadow_bf16.data_ptr(), float(lr),
                                  float(momentum), self._stream())
