@@ -380,6 +380,11 @@ void Engine::join_tree() {
       }
       Link& up0 = links_[LK_UP];
       up0.fd = -1;
+      if (up0.rccl_link) {  // half-made upgrade: tear down before retrying
+        rccl_destroy(static_cast<RcclLink*>(up0.rccl_link));
+        up0.rccl_link = nullptr;
+        up0.rccl = false;
+      }
       // a partial snapshot may have been applied; reset to the pristine
       // pre-join state (all-zero) so the retry cannot double-count
       zero_buf(values_, n_);
@@ -668,6 +673,11 @@ void Engine::reconnect_loop() try {
     } catch (const std::exception& e) {
       ::close(fd);
       up.fd = -1;
+      if (up.rccl_link) {
+        rccl_destroy(static_cast<RcclLink*>(up.rccl_link));
+        up.rccl_link = nullptr;
+        up.rccl = false;
+      }
       set_error(std::string("rejoin handshake failed: ") + e.what());
       std::this_thread::sleep_for(std::chrono::milliseconds(250));
       target = explicit_mode ? target : root_addr_;
