@@ -27,6 +27,7 @@ SOURCES = [
     "rccl_transport.cpp",
     "hip_kernels.hip",
     "ln_kernels.hip",
+    "ce_kernels.hip",
     "bindings.cpp",
 ]
 

@@ -223,6 +223,28 @@ PYBIND11_MODULE(_core, m) {
   m.def("rccl_self_test", &rccl_self_test,
         py::call_guard<py::gil_scoped_release>());
 
+  // fused bf16 cross-entropy
+  m.def("ce_fwd", [](uintptr_t logits, uintptr_t targets, uintptr_t loss,
+                     uintptr_t row_m, uintptr_t row_lse, int64_t R, int64_t V,
+                     uintptr_t stream) {
+    hip_ce_fwd(reinterpret_cast<const void*>(logits),
+               reinterpret_cast<const int32_t*>(targets),
+               reinterpret_cast<float*>(loss), reinterpret_cast<float*>(row_m),
+               reinterpret_cast<float*>(row_lse), R, V,
+               reinterpret_cast<hipStream_t>(stream));
+  });
+  m.def("ce_bwd", [](uintptr_t logits, uintptr_t targets, uintptr_t row_lse,
+                     uintptr_t dlogits, uintptr_t gscale_dev, double inv_r,
+                     int64_t R, int64_t V, uintptr_t stream) {
+    hip_ce_bwd(reinterpret_cast<const void*>(logits),
+               reinterpret_cast<const int32_t*>(targets),
+               reinterpret_cast<const float*>(row_lse),
+               reinterpret_cast<void*>(dlogits),
+               reinterpret_cast<const float*>(gscale_dev),
+               static_cast<float>(inv_r), R, V,
+               reinterpret_cast<hipStream_t>(stream));
+  });
+
   // fused bf16 LayerNorm (pointers are bf16 unless named mean/rstd/dgamma/
   // dbeta, which are fp32)
   m.def("ln_fwd", [](uintptr_t x, uintptr_t w, uintptr_t b, uintptr_t y,

@@ -92,4 +92,13 @@ void hip_ln_bwd(const void* dy, const void* x, const void* w,
                 const float* mean, const float* rstd, void* dx, float* dgamma,
                 float* dbeta, int64_t R, int C, hipStream_t s);
 
+// Fused bf16 cross-entropy (ce_kernels.hip): online-logsumexp fwd saving
+// per-row (max, lse); bwd writes bf16 dlogits = g*(softmax - onehot).
+void hip_ce_fwd(const void* logits, const int32_t* targets, float* loss,
+                float* row_m, float* row_lse, int64_t R, int64_t V,
+                hipStream_t s);
+void hip_ce_bwd(const void* logits, const int32_t* targets,
+                const float* row_lse, void* dlogits, const float* gscale_dev,
+                float inv_r, int64_t R, int64_t V, hipStream_t s);
+
 }  // namespace shamd

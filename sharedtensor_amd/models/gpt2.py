@@ -139,8 +139,13 @@ class GPT2(nn.Module):
         logits = self.lm_head(x)
         if targets is None:
             return logits, None
-        loss = F.cross_entropy(logits.view(-1, logits.size(-1)),
-                               targets.reshape(-1))
+        flat = logits.view(-1, logits.size(-1))
+        tgt = targets.reshape(-1)
+        if flat.is_cuda and flat.dtype == torch.bfloat16:
+            from ..ops import fused_ce
+            if fused_ce.can_use(flat):
+                return logits, fused_ce.fused_cross_entropy(flat, tgt)
+        loss = F.cross_entropy(flat, tgt)
         return logits, loss
 
     def num_params(self) -> int:
