@@ -36,9 +36,17 @@ static __device__ __forceinline__ uint16_t ce_f32_to_bf16(float f) {
 
 constexpr int CE_BLOCK = 256;  // 4 waves
 
-// combine two online-logsumexp states (m, s): s is sum(exp(x - m))
+// combine two online-logsumexp states (m, s): s is sum(exp(x - m)).
+// Empty states are (m = -inf, s = 0) — threads that owned no elements
+// (V < blockDim) — and must merge as identities, not as exp(-inf + inf).
 static __device__ __forceinline__ void lse_combine(float& m, float& s,
                                                    float m2, float s2) {
+  if (m2 == -INFINITY) return;
+  if (m == -INFINITY) {
+    m = m2;
+    s = s2;
+    return;
+  }
   float mn = fmaxf(m, m2);
   s = s * expf(m - mn) + s2 * expf(m2 - mn);
   m = mn;
