@@ -1,5 +1,6 @@
 """Robustness: non-finite inputs and shutdown under load."""
 import multiprocessing as mp
+import os
 import threading
 import time
 
@@ -73,3 +74,14 @@ def test_close_during_active_gossip():
         p.join(timeout=30)
         master.close()
     assert p.exitcode == 0
+
+
+def test_many_engine_lifecycles_no_leak():
+    """Open/close many engines in one process: no fd/thread leaks."""
+    port = free_port()
+    for i in range(25):
+        with st.create_or_fetch("127.0.0.1", port, torch.randn(256)) as h:
+            h.add_from_tensor(torch.ones(256))
+    # fd count should stay bounded (threads joined, sockets closed)
+    nfds = len(os.listdir("/proc/self/fd"))
+    assert nfds < 128, nfds
