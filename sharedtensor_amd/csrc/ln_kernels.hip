@@ -39,6 +39,17 @@ static __device__ __forceinline__ uint16_t ln_f32_to_bf16(float f) {
   return static_cast<uint16_t>(u >> 16);
 }
 
+static __device__ __forceinline__ float bf16pair_lo(uint32_t p) {
+  return __uint_as_float(p << 16);
+}
+static __device__ __forceinline__ float bf16pair_hi(uint32_t p) {
+  return __uint_as_float(p & 0xFFFF0000u);
+}
+static __device__ __forceinline__ uint32_t bf16pair_pack(float lo, float hi) {
+  return static_cast<uint32_t>(ln_f32_to_bf16(lo)) |
+         (static_cast<uint32_t>(ln_f32_to_bf16(hi)) << 16);
+}
+
 constexpr int LN_BLOCK = 256;  // 4 waves
 
 // Block-wide sum of one value per thread (4-wave block): wave shfl tree +
@@ -58,7 +69,7 @@ static __device__ __forceinline__ float block_sum(float v, float* lds4) {
 // in registers — a runtime loop bound would push them to scratch memory.
 constexpr int LN_MAXC = 4096;
 
-template <int CPT>
+template <int CPT>  // pairs per thread
 __global__ void k_ln_fwd(const uint16_t* __restrict__ x,
                          const uint16_t* __restrict__ w,
                          const uint16_t* __restrict__ b,
@@ -66,17 +77,26 @@ __global__ void k_ln_fwd(const uint16_t* __restrict__ x,
                          float* __restrict__ rstd, int C, float eps) {
   __shared__ float lds4[4];
   const int64_t r = blockIdx.x;
-  const uint16_t* xr = x + r * C;
-  uint16_t* yr = y + r * C;
-  float xv[CPT];
+  const int C2 = C >> 1;  // wrapper guarantees C is even
+  const uint32_t* xr = reinterpret_cast<const uint32_t*>(x + r * C);
+  const uint32_t* wp = reinterpret_cast<const uint32_t*>(w);
+  const uint32_t* bp = reinterpret_cast<const uint32_t*>(b);
+  uint32_t* yr = reinterpret_cast<uint32_t*>(y + r * C);
+  float x0[CPT], x1[CPT];
   float s = 0.f, ss = 0.f;
 #pragma unroll
   for (int k = 0; k < CPT; ++k) {
     int c = threadIdx.x + k * LN_BLOCK;
-    float v = c < C ? ln_bf16_to_f32(xr[c]) : 0.f;
-    xv[k] = v;
-    s += v;
-    ss += v * v;
+    float a = 0.f, bb = 0.f;
+    if (c < C2) {
+      uint32_t pk = xr[c];
+      a = bf16pair_lo(pk);
+      bb = bf16pair_hi(pk);
+    }
+    x0[k] = a;
+    x1[k] = bb;
+    s += a + bb;
+    ss += a * a + bb * bb;
   }
   float tot = block_sum(s, lds4);
   float tot2 = block_sum(ss, lds4);
@@ -90,10 +110,16 @@ __global__ void k_ln_fwd(const uint16_t* __restrict__ x,
 #pragma unroll
   for (int k = 0; k < CPT; ++k) {
     int c = threadIdx.x + k * LN_BLOCK;
-    if (c < C) {
-      float v = (xv[k] - m) * rs;
-      float o = v * ln_bf16_to_f32(w[c]) + (b ? ln_bf16_to_f32(b[c]) : 0.f);
-      yr[c] = ln_f32_to_bf16(o);
+    if (c < C2) {
+      uint32_t wk = wp[c];
+      float o0 = (x0[k] - m) * rs * bf16pair_lo(wk);
+      float o1 = (x1[k] - m) * rs * bf16pair_hi(wk);
+      if (bp) {
+        uint32_t bk = bp[c];
+        o0 += bf16pair_lo(bk);
+        o1 += bf16pair_hi(bk);
+      }
+      yr[c] = bf16pair_pack(o0, o1);
     }
   }
 }
@@ -107,35 +133,44 @@ __global__ void k_ln_bwd_dx(const uint16_t* __restrict__ dy,
                             uint16_t* __restrict__ dx, int C) {
   __shared__ float lds4[4];
   const int64_t r = blockIdx.x;
-  const uint16_t* xr = x + r * C;
-  const uint16_t* dyr = dy + r * C;
-  uint16_t* dxr = dx + r * C;
+  const int C2 = C >> 1;
+  const uint32_t* xr = reinterpret_cast<const uint32_t*>(x + r * C);
+  const uint32_t* dyr = reinterpret_cast<const uint32_t*>(dy + r * C);
+  const uint32_t* wp = reinterpret_cast<const uint32_t*>(w);
+  uint32_t* dxr = reinterpret_cast<uint32_t*>(dx + r * C);
   float m = mean[r], rs = rstd[r];
-  float av[CPT], xhv[CPT];
+  float a0[CPT], a1[CPT], h0[CPT], h1[CPT];
   float s1 = 0.f, s2 = 0.f;
 #pragma unroll
   for (int k = 0; k < CPT; ++k) {
     int c = threadIdx.x + k * LN_BLOCK;
-    float a = 0.f, xh = 0.f;
-    if (c < C) {
-      a = ln_bf16_to_f32(dyr[c]) * ln_bf16_to_f32(w[c]);
-      xh = (ln_bf16_to_f32(xr[c]) - m) * rs;
+    float u0 = 0.f, u1 = 0.f, v0 = 0.f, v1 = 0.f;
+    if (c < C2) {
+      uint32_t dk = dyr[c], wk = wp[c], xk = xr[c];
+      u0 = bf16pair_lo(dk) * bf16pair_lo(wk);
+      u1 = bf16pair_hi(dk) * bf16pair_hi(wk);
+      v0 = (bf16pair_lo(xk) - m) * rs;
+      v1 = (bf16pair_hi(xk) - m) * rs;
     }
-    av[k] = a;
-    xhv[k] = xh;
-    s1 += a;
-    s2 += a * xh;
+    a0[k] = u0;
+    a1[k] = u1;
+    h0[k] = v0;
+    h1[k] = v1;
+    s1 += u0 + u1;
+    s2 += u0 * v0 + u1 * v1;
   }
   float t1 = block_sum(s1, lds4) / C;
   float t2 = block_sum(s2, lds4) / C;
 #pragma unroll
   for (int k = 0; k < CPT; ++k) {
     int c = threadIdx.x + k * LN_BLOCK;
-    if (c < C) dxr[c] = ln_f32_to_bf16((av[k] - t1 - xhv[k] * t2) * rs);
+    if (c < C2)
+      dxr[c] = bf16pair_pack((a0[k] - t1 - h0[k] * t2) * rs,
+                             (a1[k] - t1 - h1[k] * t2) * rs);
   }
 }
 
-// dgamma/dbeta: each thread owns the columns {tid, tid+256, ...} and
+// dgamma/dbeta: each thread owns column pairs {tid, tid+256, ...} and
 // accumulates them in registers across its grid-stride rows; one fp32
 // atomic per owned column per block at the end.
 template <int CPT>
@@ -145,30 +180,35 @@ __global__ void k_ln_bwd_dwdb(const uint16_t* __restrict__ dy,
                               const float* __restrict__ rstd,
                               float* __restrict__ dgamma,
                               float* __restrict__ dbeta, int64_t R, int C) {
-  float accg[CPT], accb[CPT];
+  const int C2 = C >> 1;
+  float ag0[CPT], ag1[CPT], ab0[CPT], ab1[CPT];
 #pragma unroll
-  for (int k = 0; k < CPT; ++k) accg[k] = accb[k] = 0.f;
+  for (int k = 0; k < CPT; ++k) ag0[k] = ag1[k] = ab0[k] = ab1[k] = 0.f;
   for (int64_t r = blockIdx.x; r < R; r += gridDim.x) {
-    const uint16_t* xr = x + r * C;
-    const uint16_t* dyr = dy + r * C;
+    const uint32_t* xr = reinterpret_cast<const uint32_t*>(x + r * C);
+    const uint32_t* dyr = reinterpret_cast<const uint32_t*>(dy + r * C);
     float m = mean[r], rs = rstd[r];
 #pragma unroll
     for (int k = 0; k < CPT; ++k) {
       int c = threadIdx.x + k * LN_BLOCK;
-      if (c < C) {
-        float g = ln_bf16_to_f32(dyr[c]);
-        float xh = (ln_bf16_to_f32(xr[c]) - m) * rs;
-        accg[k] += g * xh;
-        accb[k] += g;
+      if (c < C2) {
+        uint32_t dk = dyr[c], xk = xr[c];
+        float g0 = bf16pair_lo(dk), g1 = bf16pair_hi(dk);
+        ag0[k] += g0 * (bf16pair_lo(xk) - m) * rs;
+        ag1[k] += g1 * (bf16pair_hi(xk) - m) * rs;
+        ab0[k] += g0;
+        ab1[k] += g1;
       }
     }
   }
 #pragma unroll
   for (int k = 0; k < CPT; ++k) {
     int c = threadIdx.x + k * LN_BLOCK;
-    if (c < C) {
-      atomicAdd(&dgamma[c], accg[k]);
-      atomicAdd(&dbeta[c], accb[k]);
+    if (c < C2) {
+      atomicAdd(&dgamma[2 * c], ag0[k]);
+      atomicAdd(&dgamma[2 * c + 1], ag1[k]);
+      atomicAdd(&dbeta[2 * c], ab0[k]);
+      atomicAdd(&dbeta[2 * c + 1], ab1[k]);
     }
   }
 }
@@ -176,8 +216,8 @@ __global__ void k_ln_bwd_dwdb(const uint16_t* __restrict__ dy,
 void hip_ln_fwd(const void* x, const void* w, const void* b, void* y,
                 float* mean, float* rstd, int64_t R, int C, float eps,
                 hipStream_t s) {
-  if (C > LN_MAXC) throw std::runtime_error("ln: C too large");
-  int cpt = (C + LN_BLOCK - 1) / LN_BLOCK;
+  if (C > LN_MAXC || (C & 1)) throw std::runtime_error("ln: bad C");
+  int cpt = (C / 2 + LN_BLOCK - 1) / LN_BLOCK;
   auto launch = [&](auto kern) {
     hipLaunchKernelGGL(kern, dim3(static_cast<uint32_t>(R)), dim3(LN_BLOCK), 0,
                        s, static_cast<const uint16_t*>(x),
@@ -197,8 +237,8 @@ void hip_ln_fwd(const void* x, const void* w, const void* b, void* y,
 void hip_ln_bwd(const void* dy, const void* x, const void* w,
                 const float* mean, const float* rstd, void* dx, float* dgamma,
                 float* dbeta, int64_t R, int C, hipStream_t s) {
-  if (C > LN_MAXC) throw std::runtime_error("ln: C too large");
-  int cpt = (C + LN_BLOCK - 1) / LN_BLOCK;
+  if (C > LN_MAXC || (C & 1)) throw std::runtime_error("ln: bad C");
+  int cpt = (C / 2 + LN_BLOCK - 1) / LN_BLOCK;
   auto launch_dx = [&](auto kern) {
     hipLaunchKernelGGL(kern, dim3(static_cast<uint32_t>(R)), dim3(LN_BLOCK), 0,
                        s, static_cast<const uint16_t*>(dy),

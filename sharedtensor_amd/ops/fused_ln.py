@@ -57,4 +57,5 @@ def fused_layer_norm(x: torch.Tensor, weight: torch.Tensor,
 def can_use(x: torch.Tensor, weight: torch.Tensor) -> bool:
     return (x.is_cuda and x.dtype == torch.bfloat16
             and weight.dtype == torch.bfloat16
-            and x.shape[-1] == weight.numel() and x.shape[-1] <= MAX_C)
+            and x.shape[-1] == weight.numel() and x.shape[-1] <= MAX_C
+            and x.shape[-1] % 2 == 0)
