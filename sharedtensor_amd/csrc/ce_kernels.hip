@@ -3,8 +3,9 @@
 // torch's route (autocast CE on (R=B*T, V=50257) bf16 logits) runs separate
 // softmax forward + backward kernels with fp32 intermediates — ~6% of a
 // GPT-2-small step.  Here:
-//   fwd: one online-logsumexp pass per row (block per row, fp32 math),
-//        emits per-row loss and saves (max, logsumexp) for backward
+//   fwd: block per row, independent max pass + exp-sum pass (fp32 math,
+//        fast-math exp — outputs are bf16-rounded anyway); emits per-row
+//        loss and saves (max, logsumexp) for backward
 //   bwd: one pass writing bf16 dlogits = g * (softmax - onehot)
 #include <hip/hip_runtime.h>
 
@@ -36,56 +37,45 @@ static __device__ __forceinline__ uint16_t ce_f32_to_bf16(float f) {
 
 constexpr int CE_BLOCK = 256;  // 4 waves
 
-// combine two online-logsumexp states (m, s): s is sum(exp(x - m)).
-// Empty states are (m = -inf, s = 0) — threads that owned no elements
-// (V < blockDim) — and must merge as identities, not as exp(-inf + inf).
-static __device__ __forceinline__ void lse_combine(float& m, float& s,
-                                                   float m2, float s2) {
-  if (m2 == -INFINITY) return;
-  if (m == -INFINITY) {
-    m = m2;
-    s = s2;
-    return;
-  }
-  float mn = fmaxf(m, m2);
-  s = s * expf(m - mn) + s2 * expf(m2 - mn);
-  m = mn;
+// block-wide sum (4-wave block): wave shfl tree + LDS combine
+static __device__ __forceinline__ float block_sum(float v, float* lds4) {
+  for (int w = 32; w > 0; w >>= 1) v += __shfl_down(v, w, 64);
+  int wave = threadIdx.x >> 6;
+  if ((threadIdx.x & 63) == 0) lds4[wave] = v;
+  __syncthreads();
+  float t = lds4[0] + lds4[1] + lds4[2] + lds4[3];
+  __syncthreads();
+  return t;
 }
 
 __global__ void k_ce_fwd(const uint16_t* __restrict__ logits,
                          const int32_t* __restrict__ targets,
                          float* __restrict__ loss, float* __restrict__ row_m,
                          float* __restrict__ row_lse, int64_t V) {
-  __shared__ float lm[4], ls[4];
+  // two independent passes (max, then sum of exp) — an online single pass
+  // has a loop-carried (m, s) dependency per thread and measured 5x slower;
+  // the second pass re-reads the row from L2 (rows are ~100 KB)
+  __shared__ float lds4[4];
   const int64_t r = blockIdx.x;
   const uint16_t* xr = logits + r * V;
-  float m = -INFINITY, s = 0.f;
-  for (int64_t v = threadIdx.x; v < V; v += CE_BLOCK) {
-    float x = ce_bf16_to_f32(xr[v]);
-    if (x > m) {
-      s = s * expf(m - x) + 1.f;
-      m = x;
-    } else {
-      s += expf(x - m);
-    }
-  }
-  for (int w = 32; w > 0; w >>= 1) {
-    float m2 = __shfl_down(m, w, 64);
-    float s2 = __shfl_down(s, w, 64);
-    lse_combine(m, s, m2, s2);
-  }
+  float m = -INFINITY;
+  for (int64_t v = threadIdx.x; v < V; v += CE_BLOCK)
+    m = fmaxf(m, ce_bf16_to_f32(xr[v]));
+  for (int w = 32; w > 0; w >>= 1) m = fmaxf(m, __shfl_down(m, w, 64));
   int wave = threadIdx.x >> 6;
-  if ((threadIdx.x & 63) == 0) {
-    lm[wave] = m;
-    ls[wave] = s;
-  }
+  if ((threadIdx.x & 63) == 0) lds4[wave] = m;
   __syncthreads();
+  m = fmaxf(fmaxf(lds4[0], lds4[1]), fmaxf(lds4[2], lds4[3]));
+  __syncthreads();
+  float s = 0.f;
+  for (int64_t v = threadIdx.x; v < V; v += CE_BLOCK)
+    s += __expf(ce_bf16_to_f32(xr[v]) - m);
+  float tot = block_sum(s, lds4);
   if (threadIdx.x == 0) {
-    for (int w = 1; w < 4; ++w) lse_combine(lm[0], ls[0], lm[w], ls[w]);
-    float lse = logf(ls[0]) + lm[0];
+    float lse = __logf(tot) + m;
     float xt = ce_bf16_to_f32(xr[targets[r]]);
     loss[r] = lse - xt;
-    row_m[r] = lm[0];
+    row_m[r] = m;
     row_lse[r] = lse;
   }
 }
@@ -104,7 +94,7 @@ __global__ void k_ce_bwd(const uint16_t* __restrict__ logits,
   float lse = row_lse[r];
   int32_t tgt = targets[r];
   for (int64_t v = threadIdx.x; v < V; v += CE_BLOCK) {
-    float p = expf(ce_bf16_to_f32(xr[v]) - lse);  // softmax
+    float p = __expf(ce_bf16_to_f32(xr[v]) - lse);  // softmax
     float g = gscale * (p - (v == tgt ? 1.f : 0.f));
     dr[v] = ce_f32_to_bf16(g);
   }

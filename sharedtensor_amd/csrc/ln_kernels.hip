@@ -170,9 +170,11 @@ __global__ void k_ln_bwd_dx(const uint16_t* __restrict__ dy,
   }
 }
 
-// dgamma/dbeta: each thread owns column pairs {tid, tid+256, ...} and
+// dgamma/dbeta: each thread owns the columns {tid, tid+256, ...} and
 // accumulates them in registers across its grid-stride rows; one fp32
-// atomic per owned column per block at the end.
+// atomic per owned column per block at the end.  (Scalar ownership: the
+// pair-vectorized variant measured 2x slower here — partial-wave idling in
+// the odd half-iteration outweighs the wider loads.)
 template <int CPT>
 __global__ void k_ln_bwd_dwdb(const uint16_t* __restrict__ dy,
                               const uint16_t* __restrict__ x,
@@ -180,35 +182,30 @@ __global__ void k_ln_bwd_dwdb(const uint16_t* __restrict__ dy,
                               const float* __restrict__ rstd,
                               float* __restrict__ dgamma,
                               float* __restrict__ dbeta, int64_t R, int C) {
-  const int C2 = C >> 1;
-  float ag0[CPT], ag1[CPT], ab0[CPT], ab1[CPT];
+  float accg[CPT], accb[CPT];
 #pragma unroll
-  for (int k = 0; k < CPT; ++k) ag0[k] = ag1[k] = ab0[k] = ab1[k] = 0.f;
+  for (int k = 0; k < CPT; ++k) accg[k] = accb[k] = 0.f;
   for (int64_t r = blockIdx.x; r < R; r += gridDim.x) {
-    const uint32_t* xr = reinterpret_cast<const uint32_t*>(x + r * C);
-    const uint32_t* dyr = reinterpret_cast<const uint32_t*>(dy + r * C);
+    const uint16_t* xr = x + r * C;
+    const uint16_t* dyr = dy + r * C;
     float m = mean[r], rs = rstd[r];
 #pragma unroll
     for (int k = 0; k < CPT; ++k) {
       int c = threadIdx.x + k * LN_BLOCK;
-      if (c < C2) {
-        uint32_t dk = dyr[c], xk = xr[c];
-        float g0 = bf16pair_lo(dk), g1 = bf16pair_hi(dk);
-        ag0[k] += g0 * (bf16pair_lo(xk) - m) * rs;
-        ag1[k] += g1 * (bf16pair_hi(xk) - m) * rs;
-        ab0[k] += g0;
-        ab1[k] += g1;
+      if (c < C) {
+        float g = ln_bf16_to_f32(dyr[c]);
+        float xh = (ln_bf16_to_f32(xr[c]) - m) * rs;
+        accg[k] += g * xh;
+        accb[k] += g;
       }
     }
   }
 #pragma unroll
   for (int k = 0; k < CPT; ++k) {
     int c = threadIdx.x + k * LN_BLOCK;
-    if (c < C2) {
-      atomicAdd(&dgamma[2 * c], ag0[k]);
-      atomicAdd(&dgamma[2 * c + 1], ag1[k]);
-      atomicAdd(&dbeta[2 * c], ab0[k]);
-      atomicAdd(&dbeta[2 * c + 1], ab1[k]);
+    if (c < C) {
+      atomicAdd(&dgamma[c], accg[k]);
+      atomicAdd(&dbeta[c], accb[k]);
     }
   }
 }
@@ -257,17 +254,18 @@ void hip_ln_bwd(const void* dy, const void* x, const void* w,
   // (256 blocks measured 0.67 TB/s — latency-bound at 1 block/CU); the
   // per-column atomic count stays trivial (2048 per column).
   int g = R < 2048 ? static_cast<int>(R) : 2048;
+  int ecpt = (C + LN_BLOCK - 1) / LN_BLOCK;  // element (not pair) ownership
   auto launch_dw = [&](auto kern) {
     hipLaunchKernelGGL(kern, dim3(g), dim3(LN_BLOCK), 0, s,
                        static_cast<const uint16_t*>(dy),
                        static_cast<const uint16_t*>(x), mean, rstd, dgamma,
                        dbeta, R, C);
   };
-  if (cpt <= 1) launch_dw(k_ln_bwd_dwdb<1>);
-  else if (cpt <= 2) launch_dw(k_ln_bwd_dwdb<2>);
-  else if (cpt <= 3) launch_dw(k_ln_bwd_dwdb<3>);
-  else if (cpt <= 4) launch_dw(k_ln_bwd_dwdb<4>);
-  else if (cpt <= 8) launch_dw(k_ln_bwd_dwdb<8>);
+  if (ecpt <= 1) launch_dw(k_ln_bwd_dwdb<1>);
+  else if (ecpt <= 2) launch_dw(k_ln_bwd_dwdb<2>);
+  else if (ecpt <= 3) launch_dw(k_ln_bwd_dwdb<3>);
+  else if (ecpt <= 4) launch_dw(k_ln_bwd_dwdb<4>);
+  else if (ecpt <= 8) launch_dw(k_ln_bwd_dwdb<8>);
   else launch_dw(k_ln_bwd_dwdb<16>);
   HIP_CHECK_LN(hipGetLastError());
 }
