@@ -1,0 +1,84 @@
+#!/usr/bin/env python3
+"""Microbenchmark the model-side fused kernels (LN, CE) vs torch equivalents
+on the GPT-2-small B=64 shapes."""
+import json
+import sys
+import time
+
+import torch
+
+sys.path.insert(0, __file__.rsplit("/", 2)[0])
+from sharedtensor_amd import _core  # noqa: E402
+
+
+def t(fn, iters=10, warmup=3):
+    for _ in range(warmup):
+        fn()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        fn()
+    torch.cuda.synchronize()
+    return (time.perf_counter() - t0) / iters * 1e3  # ms
+
+
+def main():
+    torch.cuda.set_device(0)
+    s = torch.cuda.current_stream().cuda_stream
+    out = {}
+
+    # ---- LN (R=65536, C=768) ----
+    R, C = 65536, 768
+    x = torch.randn(R, C, device="cuda").to(torch.bfloat16)
+    w = torch.randn(C, device="cuda").to(torch.bfloat16)
+    b = torch.randn(C, device="cuda").to(torch.bfloat16)
+    dy = torch.randn(R, C, device="cuda").to(torch.bfloat16)
+    y = torch.empty_like(x)
+    dx = torch.empty_like(x)
+    mean = torch.empty(R, dtype=torch.float32, device="cuda")
+    rstd = torch.empty(R, dtype=torch.float32, device="cuda")
+    dwdb = torch.zeros(2 * C, dtype=torch.float32, device="cuda")
+    out["ln_fwd_ms"] = round(t(lambda: _core.ln_fwd(
+        x.data_ptr(), w.data_ptr(), b.data_ptr(), y.data_ptr(),
+        mean.data_ptr(), rstd.data_ptr(), R, C, 1e-5, s)), 3)
+    out["ln_bwd_ms"] = round(t(lambda: _core.ln_bwd(
+        dy.data_ptr(), x.data_ptr(), w.data_ptr(), mean.data_ptr(),
+        rstd.data_ptr(), dx.data_ptr(), dwdb.data_ptr(),
+        dwdb[C:].data_ptr(), R, C, s)), 3)
+    xt = x.clone().requires_grad_(True)
+    wt = w.clone().requires_grad_(True)
+    bt = b.clone().requires_grad_(True)
+    out["torch_ln_fwd_ms"] = round(t(lambda: torch.nn.functional.layer_norm(
+        xt, (C,), wt, bt, 1e-5)), 3)
+    yt = torch.nn.functional.layer_norm(xt, (C,), wt, bt, 1e-5)
+    out["torch_ln_bwd_ms"] = round(t(lambda: torch.autograd.grad(
+        yt, (xt, wt, bt), dy, retain_graph=True)), 3)
+
+    # ---- CE (R=65536, V=50257) ----
+    R, V = 65536, 50257
+    logits = (torch.randn(R, V, device="cuda") * 2).to(torch.bfloat16)
+    targets32 = torch.randint(0, V, (R,), device="cuda", dtype=torch.int32)
+    targets = targets32.long()
+    loss = torch.empty(R, dtype=torch.float32, device="cuda")
+    row_m = torch.empty(R, dtype=torch.float32, device="cuda")
+    row_lse = torch.empty(R, dtype=torch.float32, device="cuda")
+    dlog = torch.empty_like(logits)
+    g = torch.ones((), dtype=torch.float32, device="cuda")
+    out["ce_fwd_ms"] = round(t(lambda: _core.ce_fwd(
+        logits.data_ptr(), targets32.data_ptr(), loss.data_ptr(),
+        row_m.data_ptr(), row_lse.data_ptr(), R, V, s), iters=5), 3)
+    out["ce_bwd_ms"] = round(t(lambda: _core.ce_bwd(
+        logits.data_ptr(), targets32.data_ptr(), row_lse.data_ptr(),
+        dlog.data_ptr(), g.data_ptr(), 1.0 / R, R, V, s), iters=5), 3)
+    lt = logits.clone().requires_grad_(True)
+    out["torch_ce_fwd_ms"] = round(t(lambda: torch.nn.functional.cross_entropy(
+        lt, targets), iters=5), 3)
+    lt2 = logits.clone().requires_grad_(True)
+    losst = torch.nn.functional.cross_entropy(lt2, targets)
+    out["torch_ce_bwd_ms"] = round(t(lambda: torch.autograd.grad(
+        losst, lt2, retain_graph=True), iters=5), 3)
+    print(json.dumps(out, indent=1))
+
+
+if __name__ == "__main__":
+    main()
