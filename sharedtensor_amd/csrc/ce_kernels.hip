@@ -35,6 +35,17 @@ static __device__ __forceinline__ uint16_t ce_f32_to_bf16(float f) {
   return static_cast<uint16_t>(u >> 16);
 }
 
+static __device__ __forceinline__ float ce_pair_lo(uint32_t p) {
+  return __uint_as_float(p << 16);
+}
+static __device__ __forceinline__ float ce_pair_hi(uint32_t p) {
+  return __uint_as_float(p & 0xFFFF0000u);
+}
+static __device__ __forceinline__ uint32_t ce_pair_pack(float lo, float hi) {
+  return static_cast<uint32_t>(ce_f32_to_bf16(lo)) |
+         (static_cast<uint32_t>(ce_f32_to_bf16(hi)) << 16);
+}
+
 constexpr int CE_BLOCK = 256;  // 4 waves
 
 // block-wide sum (4-wave block): wave shfl tree + LDS combine
@@ -48,6 +59,10 @@ static __device__ __forceinline__ float block_sum(float v, float* lds4) {
   return t;
 }
 
+// Rows are only 2-byte aligned when V is odd (GPT-2: V = 50257), so each
+// row is processed as [optional head elem | aligned uint32 pairs | optional
+// tail elem] — 4-byte loads halve the load-instruction count that bounds
+// these kernels.
 __global__ void k_ce_fwd(const uint16_t* __restrict__ logits,
                          const int32_t* __restrict__ targets,
                          float* __restrict__ loss, float* __restrict__ row_m,
@@ -58,18 +73,32 @@ __global__ void k_ce_fwd(const uint16_t* __restrict__ logits,
   __shared__ float lds4[4];
   const int64_t r = blockIdx.x;
   const uint16_t* xr = logits + r * V;
+  const int head = static_cast<int>(reinterpret_cast<uintptr_t>(xr) & 3) ? 1 : 0;
+  const int64_t npairs = (V - head) >> 1;
+  const bool tail = ((V - head) & 1) != 0;
+  const uint32_t* xp = reinterpret_cast<const uint32_t*>(xr + head);
+
   float m = -INFINITY;
-  for (int64_t v = threadIdx.x; v < V; v += CE_BLOCK)
-    m = fmaxf(m, ce_bf16_to_f32(xr[v]));
+  for (int64_t p = threadIdx.x; p < npairs; p += CE_BLOCK) {
+    uint32_t u = xp[p];
+    m = fmaxf(m, fmaxf(ce_pair_lo(u), ce_pair_hi(u)));
+  }
+  if (threadIdx.x == 0 && head) m = fmaxf(m, ce_bf16_to_f32(xr[0]));
+  if (threadIdx.x == 1 && tail) m = fmaxf(m, ce_bf16_to_f32(xr[V - 1]));
   for (int w = 32; w > 0; w >>= 1) m = fmaxf(m, __shfl_down(m, w, 64));
   int wave = threadIdx.x >> 6;
   if ((threadIdx.x & 63) == 0) lds4[wave] = m;
   __syncthreads();
   m = fmaxf(fmaxf(lds4[0], lds4[1]), fmaxf(lds4[2], lds4[3]));
   __syncthreads();
+
   float s = 0.f;
-  for (int64_t v = threadIdx.x; v < V; v += CE_BLOCK)
-    s += __expf(ce_bf16_to_f32(xr[v]) - m);
+  for (int64_t p = threadIdx.x; p < npairs; p += CE_BLOCK) {
+    uint32_t u = xp[p];
+    s += __expf(ce_pair_lo(u) - m) + __expf(ce_pair_hi(u) - m);
+  }
+  if (threadIdx.x == 0 && head) s += __expf(ce_bf16_to_f32(xr[0]) - m);
+  if (threadIdx.x == 1 && tail) s += __expf(ce_bf16_to_f32(xr[V - 1]) - m);
   float tot = block_sum(s, lds4);
   if (threadIdx.x == 0) {
     float lse = __logf(tot) + m;
@@ -92,11 +121,27 @@ __global__ void k_ce_bwd(const uint16_t* __restrict__ logits,
   float gscale = *gscale_dev * inv_r;  // upstream grad read on device: no
                                        // host sync in the backward pass
   float lse = row_lse[r];
-  int32_t tgt = targets[r];
-  for (int64_t v = threadIdx.x; v < V; v += CE_BLOCK) {
-    float p = __expf(ce_bf16_to_f32(xr[v]) - lse);  // softmax
-    float g = gscale * (p - (v == tgt ? 1.f : 0.f));
-    dr[v] = ce_f32_to_bf16(g);
+  int64_t tgt = targets[r];
+  const int head = static_cast<int>(reinterpret_cast<uintptr_t>(xr) & 3) ? 1 : 0;
+  const int64_t npairs = (V - head) >> 1;
+  const bool tail = ((V - head) & 1) != 0;
+  const uint32_t* xp = reinterpret_cast<const uint32_t*>(xr + head);
+  uint32_t* dp = reinterpret_cast<uint32_t*>(dr + head);
+  for (int64_t p = threadIdx.x; p < npairs; p += CE_BLOCK) {
+    uint32_t u = xp[p];
+    int64_t v0 = head + 2 * p;
+    float g0 = gscale * (__expf(ce_pair_lo(u) - lse) - (v0 == tgt ? 1.f : 0.f));
+    float g1 = gscale * (__expf(ce_pair_hi(u) - lse) - (v0 + 1 == tgt ? 1.f : 0.f));
+    dp[p] = ce_pair_pack(g0, g1);
+  }
+  if (threadIdx.x == 0 && head) {
+    float g = gscale * (__expf(ce_bf16_to_f32(xr[0]) - lse) - (0 == tgt ? 1.f : 0.f));
+    dr[0] = ce_f32_to_bf16(g);
+  }
+  if (threadIdx.x == 1 && tail) {
+    float g = gscale * (__expf(ce_bf16_to_f32(xr[V - 1]) - lse) -
+                        (V - 1 == tgt ? 1.f : 0.f));
+    dr[V - 1] = ce_f32_to_bf16(g);
   }
 }
 
