@@ -28,6 +28,7 @@ SOURCES = [
     "hip_kernels.hip",
     "ln_kernels.hip",
     "ce_kernels.hip",
+    "gelu_kernels.hip",
     "bindings.cpp",
 ]
 

@@ -223,6 +223,19 @@ PYBIND11_MODULE(_core, m) {
   m.def("rccl_self_test", &rccl_self_test,
         py::call_guard<py::gil_scoped_release>());
 
+  // fused bf16 GELU
+  m.def("gelu_fwd", [](uintptr_t x, uintptr_t y, int64_t n, uintptr_t stream) {
+    hip_gelu_fwd(reinterpret_cast<const void*>(x), reinterpret_cast<void*>(y),
+                 n, reinterpret_cast<hipStream_t>(stream));
+  });
+  m.def("gelu_bwd", [](uintptr_t dy, uintptr_t x, uintptr_t dx, int64_t n,
+                       uintptr_t stream) {
+    hip_gelu_bwd(reinterpret_cast<const void*>(dy),
+                 reinterpret_cast<const void*>(x),
+                 reinterpret_cast<void*>(dx), n,
+                 reinterpret_cast<hipStream_t>(stream));
+  });
+
   // fused bf16 cross-entropy
   m.def("ce_fwd", [](uintptr_t logits, uintptr_t targets, uintptr_t loss,
                      uintptr_t row_m, uintptr_t row_lse, int64_t R, int64_t V,

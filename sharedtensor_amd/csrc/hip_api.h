@@ -101,4 +101,10 @@ void hip_ce_bwd(const void* logits, const int32_t* targets,
                 const float* row_lse, void* dlogits, const float* gscale_dev,
                 float inv_r, int64_t R, int64_t V, hipStream_t s);
 
+// Fused bf16 GELU-tanh (gelu_kernels.hip); n must be even, buffers
+// 4-byte aligned (pair loads).
+void hip_gelu_fwd(const void* x, void* y, int64_t n, hipStream_t s);
+void hip_gelu_bwd(const void* dy, const void* x, void* dx, int64_t n,
+                  hipStream_t s);
+
 }  // namespace shamd

@@ -88,7 +88,12 @@ class MLP(nn.Module):
         self.proj = nn.Linear(4 * cfg.n_embd, cfg.n_embd)
 
     def forward(self, x):
-        return self.proj(F.gelu(self.fc(x), approximate="tanh"))
+        h = self.fc(x)
+        if h.is_cuda and h.dtype == torch.bfloat16:
+            from ..ops import fused_gelu
+            if fused_gelu.can_use(h):
+                return self.proj(fused_gelu.fused_gelu(h))
+        return self.proj(F.gelu(h, approximate="tanh"))
 
 
 class Block(nn.Module):

@@ -1,0 +1,35 @@
+"""Fused bf16 GELU-tanh (hand-written CDNA4 kernels, csrc/gelu_kernels.hip):
+uint32 pair loads/stores + fast-math tanh."""
+from __future__ import annotations
+
+import torch
+
+from .. import _core
+
+
+class _FusedGeluFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        x = x.contiguous()
+        y = torch.empty_like(x)
+        s = torch.cuda.current_stream(x.device).cuda_stream
+        _core.gelu_fwd(x.data_ptr(), y.data_ptr(), x.numel(), s)
+        ctx.save_for_backward(x)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (x,) = ctx.saved_tensors
+        dy = dy.contiguous()
+        dx = torch.empty_like(x)
+        s = torch.cuda.current_stream(x.device).cuda_stream
+        _core.gelu_bwd(dy.data_ptr(), x.data_ptr(), dx.data_ptr(), x.numel(), s)
+        return dx
+
+
+def fused_gelu(x: torch.Tensor) -> torch.Tensor:
+    return _FusedGeluFn.apply(x)
+
+
+def can_use(x: torch.Tensor) -> bool:
+    return x.is_cuda and x.dtype == torch.bfloat16 and x.numel() % 2 == 0
