@@ -54,6 +54,23 @@ def main():
     out["torch_ln_bwd_ms"] = round(t(lambda: torch.autograd.grad(
         yt, (xt, wt, bt), dy, retain_graph=True)), 3)
 
+    # ---- GELU (65536 x 3072) ----
+    xg = torch.randn(65536, 3072, device="cuda").to(torch.bfloat16)
+    yg = torch.empty_like(xg)
+    dyg = torch.randn_like(xg)
+    dxg = torch.empty_like(xg)
+    ng = xg.numel()
+    out["gelu_fwd_ms"] = round(t(lambda: _core.gelu_fwd(
+        xg.data_ptr(), yg.data_ptr(), ng, s)), 3)
+    out["gelu_bwd_ms"] = round(t(lambda: _core.gelu_bwd(
+        dyg.data_ptr(), xg.data_ptr(), dxg.data_ptr(), ng, s)), 3)
+    xt2 = xg.clone().requires_grad_(True)
+    out["torch_gelu_fwd_ms"] = round(t(lambda: torch.nn.functional.gelu(
+        xt2, approximate="tanh")), 3)
+    yt2 = torch.nn.functional.gelu(xt2, approximate="tanh")
+    out["torch_gelu_bwd_ms"] = round(t(lambda: torch.autograd.grad(
+        yt2, xt2, dyg, retain_graph=True)), 3)
+
     # ---- CE (R=65536, V=50257) ----
     R, V = 65536, 50257
     logits = (torch.randn(R, V, device="cuda") * 2).to(torch.bfloat16)
