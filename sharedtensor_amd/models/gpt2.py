@@ -88,12 +88,10 @@ class MLP(nn.Module):
         self.proj = nn.Linear(4 * cfg.n_embd, cfg.n_embd)
 
     def forward(self, x):
-        h = self.fc(x)
-        if h.is_cuda and h.dtype == torch.bfloat16:
-            from ..ops import fused_gelu
-            if fused_gelu.can_use(h):
-                return self.proj(fused_gelu.fused_gelu(h))
-        return self.proj(F.gelu(h, approximate="tanh"))
+        # torch's gelu measured FASTER than our fused kernel here (0.187 vs
+        # 0.251 ms fwd on the B=64 shape; ops/fused_gelu.py kept as a
+        # measured alternative) — keep the library kernel
+        return self.proj(F.gelu(self.fc(x), approximate="tanh"))
 
 
 class Block(nn.Module):

@@ -1,5 +1,8 @@
-"""Fused bf16 GELU-tanh (hand-written CDNA4 kernels, csrc/gelu_kernels.hip):
-uint32 pair loads/stores + fast-math tanh."""
+"""Fused bf16 GELU-tanh (hand-written CDNA4 kernels, csrc/gelu_kernels.hip).
+
+Measured SLOWER than torch's gelu forward on MI355X (0.251 vs 0.187 ms at
+(65536, 3072); backward ties) — torch's 8-wide vectorized elementwise wins.
+Kept as a correct, tested alternative; NOT wired into the models."""
 from __future__ import annotations
 
 import torch
