@@ -21,7 +21,7 @@ from typing import Optional
 
 import torch
 
-from ..engine import _SharedBase, SharedTensor
+from ..engine import _SharedBase
 
 
 def tree_parent(rank: int) -> int:
