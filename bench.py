@@ -88,6 +88,27 @@ def run_train(args, rank, world, device):
             log(f"rocm flash-attention backend: {fa}")
         except Exception as e:
             log(f"fa backend {fa} unavailable: {e}")
+    blas = os.environ.get("SHTENS_BLAS", "")
+    if blas and device.type == "cuda":
+        try:
+            torch.backends.cuda.preferred_blas_library(blas)
+            log(f"blas library: {blas}")
+        except Exception as e:
+            log(f"blas {blas} unavailable: {e}")
+    sdpa = os.environ.get("SHTENS_SDPA", "")
+    if sdpa and device.type == "cuda":
+        import sharedtensor_amd.models.gpt2 as _g
+        from torch.nn.attention import SDPBackend, sdpa_kernel
+        backend = {"flash": SDPBackend.FLASH_ATTENTION,
+                   "mem_efficient": SDPBackend.EFFICIENT_ATTENTION,
+                   "math": SDPBackend.MATH}[sdpa]
+        _orig = _g.F.scaled_dot_product_attention
+
+        def _sdpa(*a, **k):
+            with sdpa_kernel(backend):
+                return _orig(*a, **k)
+        _g.F.scaled_dot_product_attention = _sdpa
+        log(f"sdpa backend: {sdpa}")
     torch.manual_seed(1234)  # same random init on every rank
     model = GPT2(cfg).to(device)
     log(f"model {args.model}: {model.num_params()/1e6:.1f}M params, device {device}")
