@@ -83,7 +83,8 @@ class CausalSelfAttention(nn.Module):
             # back when a shape is unsupported
             from torch.nn.attention import SDPBackend, sdpa_kernel
             with sdpa_kernel([SDPBackend.EFFICIENT_ATTENTION,
-                              SDPBackend.FLASH_ATTENTION, SDPBackend.MATH]):
+                              SDPBackend.FLASH_ATTENTION, SDPBackend.MATH],
+                             set_priority=True):
                 y = F.scaled_dot_product_attention(q, k, v, is_causal=True)
         else:
             y = F.scaled_dot_product_attention(q, k, v, is_causal=True)

@@ -95,7 +95,8 @@ class LlamaAttention(nn.Module):
             # covers shapes a backend rejects, e.g. GQA corner cases)
             from torch.nn.attention import SDPBackend, sdpa_kernel
             with sdpa_kernel([SDPBackend.EFFICIENT_ATTENTION,
-                              SDPBackend.FLASH_ATTENTION, SDPBackend.MATH]):
+                              SDPBackend.FLASH_ATTENTION, SDPBackend.MATH],
+                             set_priority=True):
                 y = F.scaled_dot_product_attention(
                     q, k, v, is_causal=True,
                     enable_gqa=self.n_kv != self.n_head)
