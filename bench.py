@@ -230,6 +230,11 @@ def run_table(args, rank, world, device):
         provision_up=rank > 0,
         explicit_parent=f"127.0.0.1:{port_base + tree_parent(rank)}" if rank else "",
         listen_port=port_base + rank if world > 1 else 0)
+    # the table holds its own replica; free the model to fit 8B-scale runs
+    # (2 x ~100 GB working set on one 288 GB device)
+    del model
+    if device.type == "cuda":
+        torch.cuda.empty_cache()
     dist = dist_setup(world)
     n = sh.n
     delta = torch.randn(n, dtype=torch.float32, device=device) * 0.001
