@@ -48,7 +48,7 @@ def _child_proc(port, q, codec, snapshot, delta_scale):
             h.copy_to_tensor(out)
             return torch.allclose(out, target, atol=1e-2)
 
-        ok = wait_until(converged, timeout=30)
+        ok = wait_until(converged, timeout=60)
         if not ok:
             q.put(("fail", f"child never converged; got {out.flatten()[:4]}, "
                            f"err={h.stats()['last_error']}"))
@@ -86,7 +86,7 @@ def test_two_process_convergence(codec, snapshot):
             master.copy_to_tensor(out)
             return torch.allclose(out, target, atol=1e-2)
 
-        assert wait_until(master_sees, timeout=30), \
+        assert wait_until(master_sees, timeout=60), \
             f"master never saw child delta; got {out.flatten()[:4]} stats={master.stats()}"
         s = master.stats()
         # child's +5 arrived as gossip rounds; with snapshot join the master
@@ -121,7 +121,7 @@ def _tree_node(port, rank, q):
             h.copy_to_tensor(out)
             return torch.allclose(out, expected, atol=1e-2)
 
-        ok = wait_until(conv, timeout=45)
+        ok = wait_until(conv, timeout=75)
         q.put(("ok" if ok else "fail",
                None if ok else f"rank {rank}: {out[:8]} err={h.stats()['last_error']}"))
         time.sleep(2.0)
@@ -160,7 +160,7 @@ def test_five_node_tree_join_walk():
             master.copy_to_tensor(out)
             return torch.allclose(out, expected, atol=1e-2)
 
-        assert wait_until(conv, timeout=45), f"master: {out[:8]}"
+        assert wait_until(conv, timeout=75), f"master: {out[:8]}"
         # topology sanity: master has exactly 2 active/dead children, so at
         # least one joiner was redirected
         links = master.stats()["links"]
