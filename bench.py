@@ -3,10 +3,15 @@
 engine (BASELINE.json config 3), plus a param-sync bandwidth mode.
 
 Contract (driver): `python bench.py --gpus N --steps K --warmup W` runs one
-rank per GPU (launched via torch.distributed.run for N>1; reads RANK /
-WORLD_SIZE / LOCAL_RANK from the env), does W untimed warmup steps, times
-exactly K steps bracketed by barrier + torch.cuda.synchronize on both sides,
-takes the MAX step time over ranks, and rank 0 prints ONE JSON line.
+rank per GPU.  Two launch paths, both supported:
+  * torchrun (`torch.distributed.run --nproc-per-node N`): each rank reads
+    RANK / WORLD_SIZE / LOCAL_RANK from the env.
+  * plain `python bench.py --gpus N` with no WORLD_SIZE in the env: this
+    process SELF-LAUNCHES N rank subprocesses (env RANK/LOCAL_RANK/
+    WORLD_SIZE/MASTER_*), waits for them, and forwards rank 0's JSON line.
+Each rank does W untimed warmup steps, times exactly K steps bracketed by
+barrier + torch.cuda.synchronize on both sides, takes the MAX step time over
+ranks, and rank 0 prints ONE JSON line.
 
 The metric is the whole-job aggregate tokens/s; config reports the engine's
 param-sync wire GB/s and p50 staleness (the per-round scale: every packet
@@ -41,6 +46,66 @@ def emit(result: dict):
 def log(msg):
     r = os.environ.get("RANK", "0")
     print(f"[bench r{r}] {msg}", file=sys.stderr, flush=True)
+
+
+def self_launch(args) -> int:
+    """Spawn one rank subprocess per GPU and forward rank 0's JSON line.
+
+    Used when the driver invokes `python bench.py --gpus N` as a single
+    command (no torchrun): the rendezvous env torchrun would provide is
+    synthesized here.  Returns the exit code for the parent process.
+    """
+    import signal
+    import socket
+    import subprocess
+
+    with socket.socket() as s:  # free rendezvous port for the gloo group
+        s.bind(("127.0.0.1", 0))
+        master_port = s.getsockname()[1]
+    base_env = dict(os.environ)
+    base_env.update({"MASTER_ADDR": "127.0.0.1",
+                     "MASTER_PORT": str(master_port),
+                     "WORLD_SIZE": str(args.gpus)})
+    procs = []
+    argv = [sys.executable, os.path.abspath(__file__)] + sys.argv[1:]
+    for r in range(args.gpus):
+        env = dict(base_env)
+        env["RANK"] = str(r)
+        env["LOCAL_RANK"] = str(r)
+        procs.append(subprocess.Popen(
+            argv, env=env,
+            stdout=subprocess.PIPE if r == 0 else sys.stderr,
+            start_new_session=True))
+    log(f"self-launched {args.gpus} rank processes (master port {master_port})")
+    deadline = time.time() + float(os.environ.get("SHTENS_LAUNCH_TIMEOUT", 3600))
+    rc = 0
+    try:
+        for r, p in enumerate(procs):
+            remaining = max(1.0, deadline - time.time())
+            try:
+                p.wait(timeout=remaining)
+            except subprocess.TimeoutExpired:
+                log(f"rank {r} exceeded launch timeout; killing the job")
+                rc = 124
+                break
+            if p.returncode != 0:
+                log(f"rank {r} exited with {p.returncode}")
+                rc = rc or p.returncode
+    finally:
+        for p in procs:
+            if p.poll() is None:
+                try:  # kill the whole rank's session (it may have children)
+                    os.killpg(p.pid, signal.SIGKILL)
+                except OSError:
+                    p.kill()
+    out = procs[0].stdout.read() if procs[0].stdout else b""
+    if rc == 0:
+        lines = [l for l in out.decode(errors="replace").splitlines() if l.strip()]
+        if not lines:
+            log("rank 0 produced no JSON line")
+            return 1
+        os.write(_REAL_STDOUT, (lines[-1] + "\n").encode())
+    return rc
 
 
 def dist_setup(world):
@@ -198,6 +263,13 @@ def run_train(args, rank, world, device):
             "staleness_p90": s1["staleness_p90"],
         },
     }
+    if world == 1:
+        # self-describing: a 1-rank tree has no links, so the param-sync half
+        # of the metric does not exist rather than measuring zero
+        result["config"]["paramsync"] = "n/a at n_gpus=1 (no links)"
+        for k in ("paramsync_wire_gbps", "paramsync_logical_gbps",
+                  "sync_rounds_per_s"):
+            result["config"][k] = None
     trainer.close()
     if dist is not None:
         dist.barrier()
@@ -390,12 +462,16 @@ def main():
     ap.add_argument("--device", default="auto")
     args = ap.parse_args()
 
+    if "WORLD_SIZE" not in os.environ and args.gpus > 1:
+        sys.exit(self_launch(args))
+
     rank = int(os.environ.get("RANK", 0))
     world = int(os.environ.get("WORLD_SIZE", args.gpus))
     local = int(os.environ.get("LOCAL_RANK", rank))
     if args.device == "auto":
-        device = torch.device(f"cuda:{local}") if torch.cuda.is_available() \
-            else torch.device("cpu")
+        # modulo device count so an N-rank rehearsal also runs on fewer GPUs
+        device = torch.device(f"cuda:{local % torch.cuda.device_count()}") \
+            if torch.cuda.is_available() else torch.device("cpu")
     else:
         device = torch.device(args.device)
     if device.type == "cuda":

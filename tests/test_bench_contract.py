@@ -6,13 +6,15 @@ import subprocess
 import sys
 
 
-def run_bench(*extra):
+def run_bench(*extra, port_base="53611", timeout=240):
     env = dict(os.environ)
-    env["SHTENS_PORT_BASE"] = "53611"
+    env["SHTENS_PORT_BASE"] = port_base
+    env.pop("WORLD_SIZE", None)  # exercise the driver's plain invocation
+    env.pop("RANK", None)
     out = subprocess.run(
         [sys.executable, "bench.py", "--steps", "2", "--warmup", "1",
          "--model", "tiny", "--batch", "2", "--seq", "32", *extra],
-        capture_output=True, text=True, timeout=240, env=env)
+        capture_output=True, text=True, timeout=timeout, env=env)
     assert out.returncode == 0, out.stderr[-800:]
     lines = [l for l in out.stdout.strip().splitlines() if l.strip()]
     assert len(lines) == 1, f"stdout must be ONE JSON line, got: {lines}"
@@ -35,6 +37,34 @@ def test_train_contract():
     for key in ("model", "global_batch", "seq_len", "parallelism", "codec",
                 "staleness_p50", "paramsync_wire_gbps"):
         assert key in cfg, key
+
+
+def test_n1_paramsync_self_describing():
+    d = run_bench(port_base="53631")
+    cfg = d["config"]
+    assert cfg["paramsync"] == "n/a at n_gpus=1 (no links)"
+    assert cfg["paramsync_wire_gbps"] is None
+    assert cfg["sync_rounds_per_s"] is None
+
+
+def test_self_launch_gpus2():
+    """The driver's plain `python bench.py --gpus 2` must complete unaided:
+    bench forks both ranks itself (no torchrun, no preset WORLD_SIZE)."""
+    d = run_bench("--gpus", "2", port_base="53641", timeout=420)
+    assert d["n_gpus"] == 2
+    assert d["config"]["parallelism"] == "async-dp2"
+    # with links up, the param-sync half of the metric is measured
+    assert d["config"]["paramsync_wire_gbps"] is not None
+    assert d["config"]["sync_rounds_per_s"] >= 0
+
+
+def test_self_launch_gpus8_dry():
+    """`--gpus 8` self-launch dry-run on CPU: the full 8-rank tree forms,
+    steps, and reports without hanging (pre-stages the driver's SCALE run)."""
+    d = run_bench("--gpus", "8", "--steps", "1", port_base="53651",
+                  timeout=600)
+    assert d["n_gpus"] == 8
+    assert d["config"]["parallelism"] == "async-dp8"
 
 
 def test_paramsync_contract():
