@@ -450,26 +450,92 @@ void Engine::handshake_as_child(int fd, bool rejoin) {
     }
   }
   if (rejoin && (ah.flags & ACC_SNAPSHOT)) {
-    // reconciliation on rejoin: V := S + tmp, where S is the new parent's
-    // snapshot and tmp = our unsent up-residual (preserved in up.delta).
-    // Child slots were demoted/zeroed, so the unconnected-slot invariant
-    // (slot delta == values) is rebuilt by the same adds.
-    zero_buf(values_, n_);
-    recv_snapshot(fd);  // += S into values + child slots
+    // Reconciliation on rejoin: V := S + R, where S is the new parent's
+    // snapshot and R = our unsent up-residual at the instant reconciliation
+    // begins.  A concurrent user update u (add_from/fused_sgd) must land
+    // exactly once: R is captured into a scratch
+    // fp32 buffer and {values, every provisioned delta} reset under a brief
+    // exclusive user-op lock; the snapshot then streams WITHOUT the lock
+    // (concurrent u adds consistently into values + all deltas); finally R
+    // is re-added into values + all deltas under the lock again.  End
+    // state: values = S + R + u, and every unconnected slot == values (the
+    // invariant the reference seeds at sharedtensor.c:379-381).  No update
+    // is ever lost: u landed after link-down is inside R via up.delta.
     void* fwd0 = links_[LK_LEFT].provisioned ? links_[LK_LEFT].delta : nullptr;
     void* fwd1 = links_[LK_RIGHT].provisioned ? links_[LK_RIGHT].delta : nullptr;
-    if (gpu()) {
+    float* tmp_dev = nullptr;
+    std::vector<float> tmp_host;
+    bool captured = false;
+    {
+      std::unique_lock<std::shared_mutex> ug(user_m_);
+      if (gpu()) {
+        HIP_TRY(hipSetDevice(cfg_.device));
+        HIP_TRY(hipDeviceSynchronize());  // drain in-flight user kernels
+        if (hipMalloc(&tmp_dev, n_ * 4) == hipSuccess) {
+          HIP_TRY(hipMemsetAsync(tmp_dev, 0, n_ * 4, up.s_recv));
+          // tmp (fp32) := up.delta (fp32 or bf16 residual)
+          hip_add_delta_scatter(up.delta, cfg_.delta_bf16, n_, tmp_dev,
+                                nullptr, nullptr, up.s_recv);
+          HIP_TRY(hipStreamSynchronize(up.s_recv));
+          captured = true;
+        } else {
+          (void)hipGetLastError();  // clear; fall back to full-lock mode
+          tmp_dev = nullptr;
+        }
+      } else {
+        tmp_host.resize(static_cast<size_t>(n_));
+        for (int64_t i = 0; i < n_; ++i)
+          tmp_host[i] = atomic_load_f32(fdelta(up.delta) + i);
+        captured = true;
+      }
+      if (captured) {
+        zero_delta(up.delta);
+        if (fwd0) zero_delta(fwd0);
+        if (fwd1) zero_delta(fwd1);
+        zero_buf(values_, n_);
+      }
+    }
+    auto readd = [&] {
+      std::unique_lock<std::shared_mutex> ug(user_m_);
+      if (gpu()) {
+        HIP_TRY(hipDeviceSynchronize());
+        hip_add_scatter(tmp_dev, n_, 1.0f, values_, up.delta, fwd0, fwd1,
+                        cfg_.delta_bf16, up.s_recv);
+        HIP_TRY(hipStreamSynchronize(up.s_recv));
+        (void)hipFree(tmp_dev);
+        tmp_dev = nullptr;
+      } else {
+        for (int64_t i = 0; i < n_; ++i) {
+          float v = tmp_host[i];
+          if (v == 0.0f) continue;
+          atomic_add_f32(values_ + i, v);
+          atomic_add_f32(fdelta(up.delta) + i, v);
+          if (fwd0) atomic_add_f32(fdelta(fwd0) + i, v);
+          if (fwd1) atomic_add_f32(fdelta(fwd1) + i, v);
+        }
+      }
+    };
+    if (captured) {
+      try {
+        recv_snapshot(fd);  // += S into values + child slots (no user lock)
+      } catch (...) {
+        readd();  // restore a consistent state before the retry resets it
+        throw;
+      }
+      readd();
+    } else {
+      // scratch allocation failed (100 GB-scale tensor near HBM capacity):
+      // run the whole reconciliation under the exclusive lock instead —
+      // user ops block for the snapshot duration, but stay exactly-once
+      std::unique_lock<std::shared_mutex> ug(user_m_);
+      HIP_TRY(hipDeviceSynchronize());
+      if (fwd0) zero_delta(fwd0);
+      if (fwd1) zero_delta(fwd1);
+      zero_buf(values_, n_);
+      recv_snapshot(fd);
       hip_add_delta_scatter(up.delta, cfg_.delta_bf16, n_, values_, fwd0,
                             fwd1, up.s_recv);
       HIP_TRY(hipStreamSynchronize(up.s_recv));
-    } else {
-      for (int64_t i = 0; i < n_; ++i) {
-        float v = atomic_load_f32(fdelta(up.delta) + i);
-        if (v == 0.0f) continue;
-        atomic_add_f32(values_ + i, v);
-        if (fwd0) atomic_add_f32(fdelta(fwd0) + i, v);
-        if (fwd1) atomic_add_f32(fdelta(fwd1) + i, v);
-      }
     }
   } else if (ah.flags & ACC_SNAPSHOT) {
     recv_snapshot(fd);
@@ -599,14 +665,23 @@ void Engine::reconnect_loop() try {
     if (!try_connect(target, fd, &local)) {
       if (!explicit_mode && hops == 0 && !closing_) {
         if (failover_master()) {
-          // restore the unconnected-slot invariant (slot delta == values,
+          // Restore the unconnected-slot invariant (slot delta == values,
           // sharedtensor.c:379-381 semantics): future children must receive
-          // the full inherited state
+          // the full inherited state.  slot := values (zero + add) under
+          // the exclusive user-op lock — a plain add would double-count any
+          // update that landed in the slot since drop_children zeroed it
+          // (ADVICE round 1, medium).
           void* fwd0 = links_[LK_LEFT].provisioned ? links_[LK_LEFT].delta : nullptr;
           void* fwd1 = links_[LK_RIGHT].provisioned ? links_[LK_RIGHT].delta : nullptr;
           if (fwd0 || fwd1) {
+            std::unique_lock<std::shared_mutex> ug(user_m_);
             if (gpu()) {
               HIP_TRY(hipSetDevice(cfg_.device));
+              HIP_TRY(hipDeviceSynchronize());  // drain in-flight user kernels
+            }
+            if (fwd0) zero_delta(fwd0);
+            if (fwd1) zero_delta(fwd1);
+            if (gpu()) {
               hip_add_scatter(values_, n_, 1.0f, nullptr, fwd0, fwd1, nullptr,
                               cfg_.delta_bf16, nullptr);
               HIP_TRY(hipStreamSynchronize(nullptr));
@@ -1301,6 +1376,7 @@ void Engine::notify_dirty() { notify_all_dirty(); }
 
 void Engine::add_from(uintptr_t src, int64_t n, uintptr_t stream) {
   if (n != n_) throw std::runtime_error("add_from: size mismatch");
+  std::shared_lock<std::shared_mutex> ug(user_m_);
   const float* s = reinterpret_cast<const float*>(src);
   void* d[3];
   for (int i = 0; i < 3; ++i)
@@ -1321,6 +1397,9 @@ void Engine::add_from(uintptr_t src, int64_t n, uintptr_t stream) {
 
 void Engine::copy_to(uintptr_t dst, int64_t n, uintptr_t stream) {
   if (n != n_) throw std::runtime_error("copy_to: size mismatch");
+  // shared lock: users never observe the transiently-zeroed values of a
+  // rejoin reconciliation (stale-but-consistent is the async contract)
+  std::shared_lock<std::shared_mutex> ug(user_m_);
   if (gpu()) {
     HIP_TRY(hipMemcpyAsync(reinterpret_cast<void*>(dst), values_, n_ * 4,
                            hipMemcpyDeviceToDevice,
@@ -1333,6 +1412,7 @@ void Engine::copy_to(uintptr_t dst, int64_t n, uintptr_t stream) {
 
 void Engine::fused_sgd(uintptr_t mom, uintptr_t grad, double lr,
                        double momentum, uintptr_t stream) {
+  std::shared_lock<std::shared_mutex> ug(user_m_);
   void* d[3];
   for (int i = 0; i < 3; ++i)
     d[i] = links_[i].provisioned ? links_[i].delta : nullptr;
@@ -1362,6 +1442,7 @@ void Engine::fused_sgd_bf16(uintptr_t mom, uintptr_t grad_bf16,
                             uintptr_t stream) {
   if (!gpu())
     throw std::runtime_error("fused_sgd_bf16 is a GPU-only path");
+  std::shared_lock<std::shared_mutex> ug(user_m_);
   void* d[3];
   for (int i = 0; i < 3; ++i)
     d[i] = links_[i].provisioned ? links_[i].delta : nullptr;
@@ -1394,9 +1475,11 @@ std::vector<LinkStatsSnap> Engine::link_stats() {
 }
 
 void Engine::close() {
-  if (closing_.exchange(true)) {
-    // already closing/closed; make idempotent
-  }
+  // Idempotent AND safe under concurrency: the first caller performs the
+  // teardown; later callers (destructor racing an explicit close) block on
+  // close_m_ until it is done and then return.
+  std::lock_guard<std::mutex> cg(close_m_);
+  if (closing_.exchange(true)) return;
   // tell peers we are leaving (the reference cannot do this and exit(-1)s,
   // sharedtensor.c:421-430)
   for (auto& lk : links_) {

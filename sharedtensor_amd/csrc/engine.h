@@ -105,6 +105,12 @@ class Engine {
   int listen_port_ = 0;
   sockaddr_in listen_addr_{};
   std::mutex slots_m_;  // child-slot assignment vs reconnect demotion
+  // User mutation ops (add_from / fused_sgd* / copy_to) hold this shared;
+  // rejoin reconciliation and failover invariant-restore hold it exclusive
+  // so a concurrent update cannot land once in values and twice via a
+  // captured residual.
+  std::shared_mutex user_m_;
+  std::mutex close_m_;  // serializes concurrent close() calls
   std::thread listen_thread_;
   std::thread reconnect_thread_;
   std::atomic<bool> reconnecting_{false};

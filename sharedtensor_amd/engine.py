@@ -95,8 +95,11 @@ class _SharedBase:
             delta = torch.zeros(self.n, dtype=self.delta_dtype,
                                 device=self.device)
             if self._gpu:
-                send_buf = torch.empty(msg, dtype=torch.uint8, device=self.device)
-                recv_buf = torch.empty(msg, dtype=torch.uint8, device=self.device)
+                # zeros (one-time cost): the scales-area padding beyond 4*T
+                # bytes is never written by any kernel, and uninitialized HBM
+                # must not leak onto the wire (WIRE_FORMAT.md: pad is zero)
+                send_buf = torch.zeros(msg, dtype=torch.uint8, device=self.device)
+                recv_buf = torch.zeros(msg, dtype=torch.uint8, device=self.device)
                 send_pin = torch.empty(8 + msg, dtype=torch.uint8, pin_memory=True)
                 recv_pin = torch.empty(8 + msg, dtype=torch.uint8, pin_memory=True)
                 bufs = (delta, send_buf, recv_buf, send_pin, recv_pin)

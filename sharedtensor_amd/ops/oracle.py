@@ -126,10 +126,14 @@ def decode_1bit(payload: bytes, scale: float, n: int) -> torch.Tensor:
 # ----------------------------------------------------------------- fp8 codec
 
 def fp8_scale(delta: torch.Tensor) -> float:
-    """Power-of-two scale so that max|residual|/scale <= FP8_MAX."""
+    """Power-of-two scale so that max|residual|/scale <= FP8_MAX.
+
+    Non-finite max |residual| -> 0.0 (keepalive semantics), matching
+    cpu_compute_scale (codec_cpu.cpp) and k_finalize_scales: a poisoned
+    residual is never quantized into the wire."""
     m = delta.abs().max().item()
     if m == 0.0 or math.isnan(m) or math.isinf(m):
-        return 0.0 if m == 0.0 else np.float32(pow2_ceil(3.4e38 / FP8_MAX)).item()
+        return 0.0
     return np.float32(pow2_ceil(m / FP8_MAX)).item()
 
 
