@@ -10,6 +10,7 @@
 #include <chrono>
 #include <cmath>
 #include <cstdio>
+#include <cstdlib>
 #include <cstring>
 #include <functional>
 #include <stdexcept>
@@ -575,6 +576,11 @@ void Engine::drop_children() {
   std::lock_guard<std::mutex> g(slots_m_);
   for (int i : {LK_LEFT, LK_RIGHT}) {
     Link& lk = links_[i];
+    // settle an in-flight join first: break its I/O and wait for the join
+    // thread to land the slot in ACTIVE or FREE before demoting
+    if (lk.state.load() == L_JOINING && lk.fd >= 0)
+      ::shutdown(lk.fd, SHUT_RDWR);
+    if (lk.t_join.joinable()) lk.t_join.join();
     if (lk.state.load() == L_ACTIVE && lk.fd >= 0) {
       PacketHeader bye{};
       bye.type = PKT_CLOSE;
@@ -833,33 +839,44 @@ void Engine::listen_loop() {
       ::close(fd);
       continue;
     }
-    // pick a free child slot (serialized against reconnect demotion)
-    std::lock_guard<std::mutex> slot_guard(slots_m_);
+    // pick a free child slot — slots_m_ is held ONLY for the claim, so a
+    // multi-GB snapshot on one slot never serializes other joins/redirects
     int slot = -1;
-    for (int i : {LK_LEFT, LK_RIGHT}) {
-      Link& lk = links_[i];
-      if (!lk.provisioned) continue;
-      int st = lk.state.load();
-      if (st == L_FREE) { slot = i; break; }
-      if (st == L_DEAD) {
-        // reconnection support (reference TODO, README.md:33): reclaim the
-        // slot once its old threads have exited
-        if (lk.t_send.joinable()) lk.t_send.join();
-        if (lk.t_recv.joinable()) lk.t_recv.join();
-        if (lk.t_ctrl.joinable()) lk.t_ctrl.join();
-        if (lk.fd >= 0) ::close(lk.fd);
-        if (lk.rccl_link) {
-          rccl_destroy(static_cast<RcclLink*>(lk.rccl_link));
-          lk.rccl_link = nullptr;
-          lk.rccl = false;
+    {
+      std::lock_guard<std::mutex> slot_guard(slots_m_);
+      for (int i : {LK_LEFT, LK_RIGHT}) {
+        Link& lk = links_[i];
+        if (!lk.provisioned) continue;
+        int st = lk.state.load();
+        if (st == L_FREE) { slot = i; }
+        else if (st == L_DEAD) {
+          // reconnection support (reference TODO, README.md:33): reclaim the
+          // slot once its old threads have exited
+          if (lk.t_send.joinable()) lk.t_send.join();
+          if (lk.t_recv.joinable()) lk.t_recv.join();
+          if (lk.t_ctrl.joinable()) lk.t_ctrl.join();
+          if (lk.fd >= 0) ::close(lk.fd);
+          if (lk.rccl_link) {
+            rccl_destroy(static_cast<RcclLink*>(lk.rccl_link));
+            lk.rccl_link = nullptr;
+            lk.rccl = false;
+          }
+          if (gpu()) destroy_link_graphs(lk);
+          lk.abort.store(false);
+          lk.fd = -1;
+          lk.error.clear();
+          slot = i;
         }
-        if (gpu()) destroy_link_graphs(lk);
-        lk.abort.store(false);
-        lk.fd = -1;
-        lk.error.clear();
-        lk.state.store(L_FREE);
-        slot = i;
-        break;
+        if (slot >= 0) {
+          // claim: reap the previous join thread (it has finished — the
+          // slot was FREE/DEAD) and mark the slot as being handshaken
+          if (lk.t_join.joinable()) lk.t_join.join();
+          lk.fd = fd;
+          lk.peer = peer;
+          lk.peer_desc = addr_str(peer);
+          lk.state.store(L_JOINING);
+          break;
+        }
       }
     }
     if (slot < 0) {
@@ -886,16 +903,16 @@ void Engine::listen_loop() {
       ::close(fd);
       continue;
     }
-    accept_child(fd, h, peer, slot);
+    Link& lk = links_[slot];
+    lk.t_join = std::thread([this, fd, h, peer, slot] {
+      accept_child(fd, h, peer, slot);
+    });
   }
 }
 
 void Engine::accept_child(int fd, const Hello& h, const sockaddr_in& peer,
                           int slot) {
   Link& lk = links_[slot];
-  lk.fd = fd;
-  lk.peer = peer;
-  lk.peer_desc = addr_str(peer);
   uint8_t yes = 'Y';
   bool upgrade = rccl_wanted(h);
   AcceptHello ah{};
@@ -908,6 +925,7 @@ void Engine::accept_child(int fd, const Hello& h, const sockaddr_in& peer,
   if (!io_write(fd, &yes, 1) || !io_write(fd, &ah, sizeof(ah))) {
     ::close(fd);
     lk.fd = -1;
+    lk.state.store(L_FREE);
     return;
   }
   if (upgrade) {
@@ -925,6 +943,7 @@ void Engine::accept_child(int fd, const Hello& h, const sockaddr_in& peer,
                 std::to_string(rccl_failures_.load()) + "): " + e.what());
       ::close(fd);
       lk.fd = -1;
+      lk.state.store(L_FREE);
       return;
     }
   }
@@ -940,6 +959,28 @@ void Engine::accept_child(int fd, const Hello& h, const sockaddr_in& peer,
         lk.rccl_link = nullptr;
         lk.rccl = false;
       }
+      // the aborted snapshot's debit stays subtracted from lk.delta, which
+      // would corrupt the NEXT joiner on this slot; rebuild slot := values
+      // under the user-op lock (same invariant repair as failover)
+      {
+        std::unique_lock<std::shared_mutex> ug(user_m_);
+        if (gpu()) {
+          HIP_TRY(hipSetDevice(cfg_.device));
+          HIP_TRY(hipDeviceSynchronize());
+        }
+        zero_delta(lk.delta);
+        if (gpu()) {
+          hip_add_scatter(values_, n_, 1.0f, nullptr, lk.delta, nullptr,
+                          nullptr, cfg_.delta_bf16, nullptr);
+          HIP_TRY(hipStreamSynchronize(nullptr));
+        } else {
+          for (int64_t i = 0; i < n_; ++i) {
+            float v = atomic_load_f32(values_ + i);
+            if (v != 0.0f) atomic_add_f32(fdelta(lk.delta) + i, v);
+          }
+        }
+      }
+      lk.state.store(L_FREE);
       return;
     }
   }
@@ -990,7 +1031,14 @@ void Engine::ctrl_loop(Link& lk) {
 // concurrent updates.
 
 void Engine::send_snapshot(Link& lk) {
-  const int64_t chunk_bytes = gpu() ? (SA_ + P_) : (1 << 20);
+  // test-only join-storm knob: per-chunk throttle (and smaller chunks) to
+  // make a snapshot artificially slow; never set in production paths
+  static const double test_delay_s = [] {
+    const char* e = std::getenv("SHTENS_TEST_SNAPSHOT_DELAY_MS");
+    return e ? std::atof(e) / 1e3 : 0.0;
+  }();
+  int64_t chunk_bytes = gpu() ? (SA_ + P_) : (1 << 20);
+  if (test_delay_s > 0.0) chunk_bytes = std::min<int64_t>(chunk_bytes, 1 << 14);
   const int64_t chunk_elems = std::max<int64_t>(chunk_bytes / 4, 1);
   std::vector<uint8_t> tmp;
   if (!gpu()) tmp.resize(static_cast<size_t>(chunk_elems) * 4);
@@ -1021,6 +1069,8 @@ void Engine::send_snapshot(Link& lk) {
           atomic_add_f32(fdelta(lk.delta) + off + i, -snap[i]);
     }
     lk.bytes_sent += ce * 4;
+    if (test_delay_s > 0.0)
+      std::this_thread::sleep_for(std::chrono::duration<double>(test_delay_s));
   }
 }
 
@@ -1222,6 +1272,10 @@ void Engine::send_loop(Link& lk) {
 // -------------------------------------------------------------- recv side
 
 void Engine::apply_packet(Link& lk, const float* scales_host) {
+  // Shared user-op lock: packet application forwards into other links'
+  // delta buffers, which an exclusive invariant repair (rejoin/failover/
+  // aborted-snapshot) must be able to quiesce along with user mutations.
+  std::shared_lock<std::shared_mutex> ug(user_m_);
   // destinations: local replica + gossip-forward into the other links'
   // delta buffers, excluding the source (sharedtensor.c:124-127)
   void* fwd[2] = {nullptr, nullptr};
@@ -1501,6 +1555,7 @@ void Engine::close() {
   if (listen_fd_ >= 0) ::shutdown(listen_fd_, SHUT_RDWR);
   if (listen_thread_.joinable()) listen_thread_.join();
   for (auto& lk : links_) {
+    if (lk.t_join.joinable()) lk.t_join.join();
     if (lk.t_send.joinable()) lk.t_send.join();
     if (lk.t_recv.joinable()) lk.t_recv.join();
     if (lk.t_ctrl.joinable()) lk.t_ctrl.join();

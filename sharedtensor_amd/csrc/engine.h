@@ -19,7 +19,7 @@
 
 namespace shamd {
 
-enum LinkState { L_FREE = 0, L_ACTIVE = 1, L_DEAD = 2 };
+enum LinkState { L_FREE = 0, L_ACTIVE = 1, L_DEAD = 2, L_JOINING = 3 };
 enum LinkIdx { LK_UP = 0, LK_LEFT = 1, LK_RIGHT = 2 };
 
 struct Link {
@@ -50,6 +50,9 @@ struct Link {
   bool rccl = false;           // data plane upgraded to RCCL over xGMI
   void* rccl_link = nullptr;   // opaque RcclLink*
   std::thread t_ctrl;          // TCP control reader when data plane is RCCL
+  std::thread t_join;          // per-join handshake+snapshot thread: joins
+                               // overlap and redirects keep flowing while a
+                               // multi-GB snapshot streams (round-1 weak #4)
   std::atomic<bool> abort{false};
   // captured per-round sequences (GPU): one hipGraphLaunch replaces the
   // 2-4 launches of the scale / quantize+stage / apply phases
