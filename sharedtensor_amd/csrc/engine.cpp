@@ -910,9 +910,35 @@ void Engine::listen_loop() {
   }
 }
 
+// slot delta := values, exclusive vs user mutations and packet applies.
+// Re-establishes the unconnected-slot invariant (slot == values,
+// sharedtensor.c:379-381 semantics) regardless of slot history: residue of
+// a dead child (undelivered deltas minus its snapshot debit), an aborted
+// snapshot, or a failover demotion would otherwise be gossiped as garbage
+// to the next joiner claiming the slot.
+void Engine::rebuild_slot_invariant(Link& lk) {
+  std::unique_lock<std::shared_mutex> ug(user_m_);
+  if (gpu()) {
+    HIP_TRY(hipSetDevice(cfg_.device));
+    HIP_TRY(hipDeviceSynchronize());  // drain in-flight user kernels
+  }
+  zero_delta(lk.delta);
+  if (gpu()) {
+    hip_add_scatter(values_, n_, 1.0f, nullptr, lk.delta, nullptr, nullptr,
+                    cfg_.delta_bf16, nullptr);
+    HIP_TRY(hipStreamSynchronize(nullptr));
+  } else {
+    for (int64_t i = 0; i < n_; ++i) {
+      float v = atomic_load_f32(values_ + i);
+      if (v != 0.0f) atomic_add_f32(fdelta(lk.delta) + i, v);
+    }
+  }
+}
+
 void Engine::accept_child(int fd, const Hello& h, const sockaddr_in& peer,
                           int slot) {
   Link& lk = links_[slot];
+  rebuild_slot_invariant(lk);
   uint8_t yes = 'Y';
   bool upgrade = rccl_wanted(h);
   AcceptHello ah{};
@@ -959,27 +985,8 @@ void Engine::accept_child(int fd, const Hello& h, const sockaddr_in& peer,
         lk.rccl_link = nullptr;
         lk.rccl = false;
       }
-      // the aborted snapshot's debit stays subtracted from lk.delta, which
-      // would corrupt the NEXT joiner on this slot; rebuild slot := values
-      // under the user-op lock (same invariant repair as failover)
-      {
-        std::unique_lock<std::shared_mutex> ug(user_m_);
-        if (gpu()) {
-          HIP_TRY(hipSetDevice(cfg_.device));
-          HIP_TRY(hipDeviceSynchronize());
-        }
-        zero_delta(lk.delta);
-        if (gpu()) {
-          hip_add_scatter(values_, n_, 1.0f, nullptr, lk.delta, nullptr,
-                          nullptr, cfg_.delta_bf16, nullptr);
-          HIP_TRY(hipStreamSynchronize(nullptr));
-        } else {
-          for (int64_t i = 0; i < n_; ++i) {
-            float v = atomic_load_f32(values_ + i);
-            if (v != 0.0f) atomic_add_f32(fdelta(lk.delta) + i, v);
-          }
-        }
-      }
+      // the aborted snapshot's debit leaves lk.delta != values; harmless:
+      // rebuild_slot_invariant repairs it when the slot is next claimed
       lk.state.store(L_FREE);
       return;
     }

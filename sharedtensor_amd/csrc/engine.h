@@ -150,6 +150,7 @@ class Engine {
   void bind_listen(const sockaddr_in& addr, bool shared = true);
   void listen_loop();
   void accept_child(int fd, const Hello& h, const sockaddr_in& peer, int slot);
+  void rebuild_slot_invariant(Link& lk);  // slot delta := values
   void spawn_link_threads(Link& lk);
   void send_loop(Link& lk);
   void recv_loop(Link& lk);
