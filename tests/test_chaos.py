@@ -47,7 +47,7 @@ def _chaos_rank(rank, port_base, stop_ev, q):
         while not stop_ev.is_set():
             sh._add_flat(delta)
             time.sleep(0.05)
-        time.sleep(5.0)  # drain: residuals decay geometrically to ~0
+        time.sleep(8.0)  # drain: residuals decay geometrically to ~0
         st = sh.stats()
         q.put(("ok", rank, sh.values.clone(), st["reconnects"], fd0,
                _fd_count(), st["last_error"]))
@@ -79,7 +79,7 @@ def test_tree_chaos_kill_restart_interior_ranks():
             procs[victim].join(timeout=10)
             time.sleep(1.0)
             procs[victim] = spawn(victim)
-            time.sleep(7.0)  # heal window: rejoin + snapshot + drain
+            time.sleep(9.0)  # heal window: rejoin + snapshot + drain
 
         stop_ev.set()
         reports = []
