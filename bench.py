@@ -141,9 +141,15 @@ def sum_over_ranks(dist, x: float) -> float:
 
 def run_train(args, rank, world, device):
     from sharedtensor_amd.models.gpt2 import GPT2, GPT2Config
+    from sharedtensor_amd.models.llama import Llama, LlamaConfig
     from sharedtensor_amd.parallel.async_dp import AsyncDPTrainer
 
-    cfg = GPT2Config.tiny() if args.model == "tiny" else GPT2Config.small()
+    llama = args.model.startswith("llama")
+    if llama:
+        cfg = {"llama1b": LlamaConfig.llama_1b,
+               "llama8b": LlamaConfig.llama3_8b}[args.model]()
+    else:
+        cfg = GPT2Config.tiny() if args.model == "tiny" else GPT2Config.small()
     if args.seq:
         cfg.block_size = min(cfg.block_size, args.seq) if args.model == "tiny" else args.seq
     fa = os.environ.get("SHTENS_FA", "")
@@ -175,7 +181,7 @@ def run_train(args, rank, world, device):
         _g.F.scaled_dot_product_attention = _sdpa
         log(f"sdpa backend: {sdpa}")
     torch.manual_seed(1234)  # same random init on every rank
-    model = GPT2(cfg).to(device)
+    model = (Llama(cfg) if llama else GPT2(cfg)).to(device)
     log(f"model {args.model}: {model.num_params()/1e6:.1f}M params, device {device}")
 
     use_bf16_params = device.type == "cuda" and not args.fp32_params

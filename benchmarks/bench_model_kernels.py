@@ -71,6 +71,34 @@ def main():
     out["torch_gelu_bwd_ms"] = round(t(lambda: torch.autograd.grad(
         yt2, xt2, dyg, retain_graph=True)), 3)
 
+    # ---- RMSNorm (llama-1B shape R=B*T=16384, C=2048; 8B C=4096) ----
+    for C2 in (2048, 4096):
+        R2 = 16384
+        xr = torch.randn(R2, C2, device="cuda").to(torch.bfloat16)
+        wr = torch.randn(C2, device="cuda").to(torch.bfloat16)
+        dyr = torch.randn(R2, C2, device="cuda").to(torch.bfloat16)
+        yr = torch.empty_like(xr)
+        dxr = torch.empty_like(xr)
+        rstd2 = torch.empty(R2, dtype=torch.float32, device="cuda")
+        dg = torch.zeros(C2, dtype=torch.float32, device="cuda")
+        out[f"rms{C2}_fwd_ms"] = round(t(lambda: _core.rms_fwd(
+            xr.data_ptr(), wr.data_ptr(), yr.data_ptr(), rstd2.data_ptr(),
+            R2, C2, 1e-5, s)), 3)
+        out[f"rms{C2}_bwd_ms"] = round(t(lambda: _core.rms_bwd(
+            dyr.data_ptr(), xr.data_ptr(), wr.data_ptr(), rstd2.data_ptr(),
+            dxr.data_ptr(), dg.data_ptr(), R2, C2, s)), 3)
+
+        def torch_rms(xi, wi):  # the pre-fusion module path (fp32 upcast)
+            xf = xi.float()
+            h = xf * torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + 1e-5)
+            return (h * wi.float()).to(xi.dtype)
+        xt3 = xr.clone().requires_grad_(True)
+        wt3 = wr.clone().requires_grad_(True)
+        out[f"torch_rms{C2}_fwd_ms"] = round(t(lambda: torch_rms(xt3, wt3)), 3)
+        yt3 = torch_rms(xt3, wt3)
+        out[f"torch_rms{C2}_bwd_ms"] = round(t(lambda: torch.autograd.grad(
+            yt3, (xt3, wt3), dyr, retain_graph=True)), 3)
+
     # ---- CE (R=65536, V=50257) ----
     R, V = 65536, 50257
     logits = (torch.randn(R, V, device="cuda") * 2).to(torch.bfloat16)

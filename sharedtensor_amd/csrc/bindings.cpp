@@ -270,6 +270,23 @@ PYBIND11_MODULE(_core, m) {
                R, C, static_cast<float>(eps),
                reinterpret_cast<hipStream_t>(stream));
   });
+  m.def("rms_fwd", [](uintptr_t x, uintptr_t w, uintptr_t y, uintptr_t rstd,
+                      int64_t R, int C, double eps, uintptr_t stream) {
+    hip_rms_fwd(reinterpret_cast<const void*>(x),
+                reinterpret_cast<const void*>(w), reinterpret_cast<void*>(y),
+                reinterpret_cast<float*>(rstd), R, C, static_cast<float>(eps),
+                reinterpret_cast<hipStream_t>(stream));
+  });
+  m.def("rms_bwd", [](uintptr_t dy, uintptr_t x, uintptr_t w, uintptr_t rstd,
+                      uintptr_t dx, uintptr_t dgamma, int64_t R, int C,
+                      uintptr_t stream) {
+    hip_rms_bwd(reinterpret_cast<const void*>(dy),
+                reinterpret_cast<const void*>(x),
+                reinterpret_cast<const void*>(w),
+                reinterpret_cast<const float*>(rstd),
+                reinterpret_cast<void*>(dx), reinterpret_cast<float*>(dgamma),
+                R, C, reinterpret_cast<hipStream_t>(stream));
+  });
   m.def("ln_bwd", [](uintptr_t dy, uintptr_t x, uintptr_t w, uintptr_t mean,
                      uintptr_t rstd, uintptr_t dx, uintptr_t dgamma,
                      uintptr_t dbeta, int64_t R, int C, uintptr_t stream) {

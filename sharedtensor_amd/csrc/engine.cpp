@@ -1002,9 +1002,18 @@ void Engine::spawn_link_threads(Link& lk) {
 }
 
 bool Engine::rccl_wanted(const Hello& h) const {
+  // Test-only override: let two processes sharing ONE device drive the full
+  // ncclSend/ncclRecv data plane (enqueue ordering, stream polling, abort,
+  // teardown) on a single leased GPU.  Same-device pairs are declined by
+  // default — on one device the TCP loopback path is both correct and has
+  // no xGMI to win back.
+  static const bool force_same_dev = [] {
+    const char* e = std::getenv("SHTENS_RCCL_FORCE_SAME_DEVICE");
+    return e && e[0] == '1';
+  }();
   return gpu() && cfg_.use_rccl && rccl_failures_.load() < 2 &&
          (h.flags & HELLO_WANT_RCCL) && h.hostid == hostid_ && h.device >= 0 &&
-         h.device != cfg_.device;
+         (h.device != cfg_.device || force_same_dev);
 }
 
 void Engine::rccl_upgrade(Link& lk, const uint8_t* ids, bool is_parent) {

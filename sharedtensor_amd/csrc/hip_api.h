@@ -92,6 +92,15 @@ void hip_ln_bwd(const void* dy, const void* x, const void* w,
                 const float* mean, const float* rstd, void* dx, float* dgamma,
                 float* dbeta, int64_t R, int C, hipStream_t s);
 
+// Fused bf16 RMSNorm for the Llama training path (ln_kernels.hip): fwd
+// saves fp32 rstd; bwd = single-reduction dx pass + register-accumulated
+// dgamma pass.
+void hip_rms_fwd(const void* x, const void* w, void* y, float* rstd,
+                 int64_t R, int C, float eps, hipStream_t s);
+void hip_rms_bwd(const void* dy, const void* x, const void* w,
+                 const float* rstd, void* dx, float* dgamma, int64_t R, int C,
+                 hipStream_t s);
+
 // Fused bf16 cross-entropy (ce_kernels.hip): online-logsumexp fwd saving
 // per-row (max, lse); bwd writes bf16 dlogits = g*(softmax - onehot).
 void hip_ce_fwd(const void* logits, const int32_t* targets, float* loss,

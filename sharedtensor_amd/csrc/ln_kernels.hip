@@ -210,6 +210,177 @@ __global__ void k_ln_bwd_dwdb(const uint16_t* __restrict__ dy,
   }
 }
 
+// ------------------------------------------------------------- RMSNorm
+// Fused bf16 RMSNorm for the Llama training path (round-1 weak #5: the
+// module-level fp32 upcast re-read the whole (B,T,C) activation per call —
+// the same cast-traffic tax FusedLayerNorm removed for GPT-2).
+//   fwd:    y = x * rsqrt(mean(x^2) + eps) * w, one pass, saves fp32 rstd
+//   bwd dx: dx = rs * (dy*w - x * rs^2 * mean(dy*w*x))  — ONE row reduction
+//   bwd dw: grid-stride rows, register column accumulators (as LN dw/db)
+
+template <int CPT>  // pairs per thread
+__global__ void k_rms_fwd(const uint16_t* __restrict__ x,
+                          const uint16_t* __restrict__ w,
+                          uint16_t* __restrict__ y, float* __restrict__ rstd,
+                          int C, float eps) {
+  __shared__ float lds4[4];
+  const int64_t r = blockIdx.x;
+  const int C2 = C >> 1;
+  const uint32_t* xr = reinterpret_cast<const uint32_t*>(x + r * C);
+  const uint32_t* wp = reinterpret_cast<const uint32_t*>(w);
+  uint32_t* yr = reinterpret_cast<uint32_t*>(y + r * C);
+  float x0[CPT], x1[CPT];
+  float ss = 0.f;
+#pragma unroll
+  for (int k = 0; k < CPT; ++k) {
+    int c = threadIdx.x + k * LN_BLOCK;
+    float a = 0.f, b = 0.f;
+    if (c < C2) {
+      uint32_t pk = xr[c];
+      a = bf16pair_lo(pk);
+      b = bf16pair_hi(pk);
+    }
+    x0[k] = a;
+    x1[k] = b;
+    ss += a * a + b * b;
+  }
+  float tot2 = block_sum(ss, lds4);
+  float rs = rsqrtf(tot2 / C + eps);
+  if (threadIdx.x == 0) rstd[r] = rs;
+#pragma unroll
+  for (int k = 0; k < CPT; ++k) {
+    int c = threadIdx.x + k * LN_BLOCK;
+    if (c < C2) {
+      uint32_t wk = wp[c];
+      yr[c] = bf16pair_pack(x0[k] * rs * bf16pair_lo(wk),
+                            x1[k] * rs * bf16pair_hi(wk));
+    }
+  }
+}
+
+template <int CPT>
+__global__ void k_rms_bwd_dx(const uint16_t* __restrict__ dy,
+                             const uint16_t* __restrict__ x,
+                             const uint16_t* __restrict__ w,
+                             const float* __restrict__ rstd,
+                             uint16_t* __restrict__ dx, int C) {
+  __shared__ float lds4[4];
+  const int64_t r = blockIdx.x;
+  const int C2 = C >> 1;
+  const uint32_t* xr = reinterpret_cast<const uint32_t*>(x + r * C);
+  const uint32_t* dyr = reinterpret_cast<const uint32_t*>(dy + r * C);
+  const uint32_t* wp = reinterpret_cast<const uint32_t*>(w);
+  uint32_t* dxr = reinterpret_cast<uint32_t*>(dx + r * C);
+  float rs = rstd[r];
+  float g0[CPT], g1[CPT], v0[CPT], v1[CPT];
+  float s = 0.f;
+#pragma unroll
+  for (int k = 0; k < CPT; ++k) {
+    int c = threadIdx.x + k * LN_BLOCK;
+    float u0 = 0.f, u1 = 0.f, a = 0.f, b = 0.f;
+    if (c < C2) {
+      uint32_t dk = dyr[c], wk = wp[c], xk = xr[c];
+      u0 = bf16pair_lo(dk) * bf16pair_lo(wk);
+      u1 = bf16pair_hi(dk) * bf16pair_hi(wk);
+      a = bf16pair_lo(xk);
+      b = bf16pair_hi(xk);
+    }
+    g0[k] = u0;
+    g1[k] = u1;
+    v0[k] = a;
+    v1[k] = b;
+    s += u0 * a + u1 * b;
+  }
+  float t = block_sum(s, lds4) / C * rs * rs;
+#pragma unroll
+  for (int k = 0; k < CPT; ++k) {
+    int c = threadIdx.x + k * LN_BLOCK;
+    if (c < C2)
+      dxr[c] = bf16pair_pack((g0[k] - v0[k] * t) * rs,
+                             (g1[k] - v1[k] * t) * rs);
+  }
+}
+
+template <int CPT>
+__global__ void k_rms_bwd_dw(const uint16_t* __restrict__ dy,
+                             const uint16_t* __restrict__ x,
+                             const float* __restrict__ rstd,
+                             float* __restrict__ dgamma, int64_t R, int C) {
+  float accg[CPT];
+#pragma unroll
+  for (int k = 0; k < CPT; ++k) accg[k] = 0.f;
+  for (int64_t r = blockIdx.x; r < R; r += gridDim.x) {
+    const uint16_t* xr = x + r * C;
+    const uint16_t* dyr = dy + r * C;
+    float rs = rstd[r];
+#pragma unroll
+    for (int k = 0; k < CPT; ++k) {
+      int c = threadIdx.x + k * LN_BLOCK;
+      if (c < C)
+        accg[k] += ln_bf16_to_f32(dyr[c]) * ln_bf16_to_f32(xr[c]) * rs;
+    }
+  }
+#pragma unroll
+  for (int k = 0; k < CPT; ++k) {
+    int c = threadIdx.x + k * LN_BLOCK;
+    if (c < C) atomicAdd(&dgamma[c], accg[k]);
+  }
+}
+
+void hip_rms_fwd(const void* x, const void* w, void* y, float* rstd,
+                 int64_t R, int C, float eps, hipStream_t s) {
+  if (C > LN_MAXC || (C & 1)) throw std::runtime_error("rms: bad C");
+  int cpt = (C / 2 + LN_BLOCK - 1) / LN_BLOCK;
+  auto launch = [&](auto kern) {
+    hipLaunchKernelGGL(kern, dim3(static_cast<uint32_t>(R)), dim3(LN_BLOCK), 0,
+                       s, static_cast<const uint16_t*>(x),
+                       static_cast<const uint16_t*>(w),
+                       static_cast<uint16_t*>(y), rstd, C, eps);
+  };
+  if (cpt <= 1) launch(k_rms_fwd<1>);
+  else if (cpt <= 2) launch(k_rms_fwd<2>);
+  else if (cpt <= 3) launch(k_rms_fwd<3>);
+  else if (cpt <= 4) launch(k_rms_fwd<4>);
+  else if (cpt <= 8) launch(k_rms_fwd<8>);
+  else launch(k_rms_fwd<16>);
+  HIP_CHECK_LN(hipGetLastError());
+}
+
+void hip_rms_bwd(const void* dy, const void* x, const void* w,
+                 const float* rstd, void* dx, float* dgamma, int64_t R, int C,
+                 hipStream_t s) {
+  if (C > LN_MAXC || (C & 1)) throw std::runtime_error("rms: bad C");
+  int cpt = (C / 2 + LN_BLOCK - 1) / LN_BLOCK;
+  auto launch_dx = [&](auto kern) {
+    hipLaunchKernelGGL(kern, dim3(static_cast<uint32_t>(R)), dim3(LN_BLOCK), 0,
+                       s, static_cast<const uint16_t*>(dy),
+                       static_cast<const uint16_t*>(x),
+                       static_cast<const uint16_t*>(w), rstd,
+                       static_cast<uint16_t*>(dx), C);
+  };
+  if (cpt <= 1) launch_dx(k_rms_bwd_dx<1>);
+  else if (cpt <= 2) launch_dx(k_rms_bwd_dx<2>);
+  else if (cpt <= 3) launch_dx(k_rms_bwd_dx<3>);
+  else if (cpt <= 4) launch_dx(k_rms_bwd_dx<4>);
+  else if (cpt <= 8) launch_dx(k_rms_bwd_dx<8>);
+  else launch_dx(k_rms_bwd_dx<16>);
+  HIP_CHECK_LN(hipGetLastError());
+  int g = R < 2048 ? static_cast<int>(R) : 2048;  // 8 blocks/CU (see LN note)
+  int ecpt = (C + LN_BLOCK - 1) / LN_BLOCK;
+  auto launch_dw = [&](auto kern) {
+    hipLaunchKernelGGL(kern, dim3(g), dim3(LN_BLOCK), 0, s,
+                       static_cast<const uint16_t*>(dy),
+                       static_cast<const uint16_t*>(x), rstd, dgamma, R, C);
+  };
+  if (ecpt <= 1) launch_dw(k_rms_bwd_dw<1>);
+  else if (ecpt <= 2) launch_dw(k_rms_bwd_dw<2>);
+  else if (ecpt <= 3) launch_dw(k_rms_bwd_dw<3>);
+  else if (ecpt <= 4) launch_dw(k_rms_bwd_dw<4>);
+  else if (ecpt <= 8) launch_dw(k_rms_bwd_dw<8>);
+  else launch_dw(k_rms_bwd_dw<16>);
+  HIP_CHECK_LN(hipGetLastError());
+}
+
 void hip_ln_fwd(const void* x, const void* w, const void* b, void* y,
                 float* mean, float* rstd, int64_t R, int C, float eps,
                 hipStream_t s) {

@@ -49,6 +49,13 @@ class RMSNorm(nn.Module):
         self.eps = eps
 
     def forward(self, x):
+        if x.is_cuda and x.dtype == torch.bfloat16 \
+                and self.weight.dtype == torch.bfloat16:
+            # fused CDNA4 kernel: no fp32 upcast traffic on the (B,T,C)
+            # activation (csrc/ln_kernels.hip, fp32 statistics inside)
+            from ..ops import fused_rms
+            if fused_rms.can_use(x, self.weight):
+                return fused_rms.fused_rms_norm(x, self.weight, self.eps)
         dt = x.dtype
         x = x.float()
         x = x * torch.rsqrt(x.pow(2).mean(-1, keepdim=True) + self.eps)
