@@ -220,6 +220,10 @@ PYBIND11_MODULE(_core, m) {
   m.def("gpu_fused_sgd", &py_gpu_fused_sgd, py::arg("mom"), py::arg("grad"),
         py::arg("lr"), py::arg("mu"), py::arg("n"), py::arg("dsts"),
         py::arg("stream"), py::arg("delta_bf16") = false);
+  m.def("rccl_loopback_payload", &rccl_loopback_payload,
+        "Self ncclSend/ncclRecv of a real payload on one device (1-rank "
+        "comm): executes the non-blocking enqueue ordering + stream polling "
+        "the 2-GPU xGMI links use and verifies the moved bytes.");
   m.def("rccl_self_test", &rccl_self_test,
         py::call_guard<py::gil_scoped_release>());
 

@@ -1017,6 +1017,13 @@ bool Engine::rccl_wanted(const Hello& h) const {
 }
 
 void Engine::rccl_upgrade(Link& lk, const uint8_t* ids, bool is_parent) {
+  // Test-only fault injection: fail the upgrade deterministically (before
+  // any comm is made) so the negotiation-failure -> TCP-fallback path can
+  // be executed on a single GPU.  A REAL same-device duplicate comm is
+  // rejected by RCCL ("Duplicate GPU detected") and aborting the half-made
+  // comm can hang, so the fallback is tested via injection instead.
+  if (std::getenv("SHTENS_TEST_RCCL_FAIL"))
+    throw std::runtime_error("injected rccl failure (SHTENS_TEST_RCCL_FAIL)");
   double to = cfg_.join_timeout_s < 20.0 ? cfg_.join_timeout_s : 20.0;
   lk.rccl_link = rccl_link_create(cfg_.device, ids, is_parent, to);
   lk.rccl = true;

@@ -7,6 +7,7 @@
 #include <stdexcept>
 #include <string>
 #include <thread>
+#include <vector>
 
 namespace shamd {
 
@@ -178,6 +179,66 @@ void rccl_self_test(int device) {
   (void)hipStreamDestroy(s);
   (void)hipFree(buf);
   (void)ncclCommDestroy(c);
+}
+
+// Move a real payload through ncclSend/ncclRecv on one device: a 1-rank
+// comm self-send/recv (grouped so the pair matches inside one kernel
+// launch).  Exercises the non-blocking enqueue ordering (wait on comm async
+// state, then the stream) with actual data movement and verifies the bytes.
+// RCCL cannot make a 2-rank comm on ONE device ("Duplicate GPU detected",
+// and aborting a half-made duplicate comm can hang), so this is the deepest
+// payload-path execution a single leased GPU allows; the 2-GPU xGMI links
+// use the identical rccl_send/rccl_recv code.
+void rccl_loopback_payload(int device, int64_t bytes) {
+  if (hipSetDevice(device) != hipSuccess)
+    throw std::runtime_error("hipSetDevice failed");
+  ncclUniqueId id;
+  NCCL_TRY(ncclGetUniqueId(&id));
+  ncclComm_t c = nullptr;
+  ncclConfig_t cfg = NCCL_CONFIG_INITIALIZER;
+  cfg.blocking = 0;
+  NCCL_TRY(ncclCommInitRankConfig(&c, 1, id, 0, &cfg));
+  wait_comm(c, 30.0, "loopback init");
+  uint8_t *src = nullptr, *dst = nullptr;
+  std::vector<uint8_t> host(static_cast<size_t>(bytes));
+  for (int64_t i = 0; i < bytes; ++i)
+    host[static_cast<size_t>(i)] = static_cast<uint8_t>(i * 131 + 7);
+  hipStream_t s = nullptr;
+  auto cleanup = [&] {
+    if (s) (void)hipStreamDestroy(s);
+    if (src) (void)hipFree(src);
+    if (dst) (void)hipFree(dst);
+    if (c) (void)ncclCommDestroy(c);
+  };
+  try {
+    if (hipMalloc(&src, bytes) != hipSuccess ||
+        hipMalloc(&dst, bytes) != hipSuccess)
+      throw std::runtime_error("hipMalloc failed");
+    if (hipMemcpy(src, host.data(), bytes, hipMemcpyHostToDevice) != hipSuccess)
+      throw std::runtime_error("H2D failed");
+    (void)hipMemset(dst, 0, bytes);
+    if (hipStreamCreateWithFlags(&s, hipStreamNonBlocking) != hipSuccess)
+      throw std::runtime_error("stream create failed");
+    NCCL_TRY(ncclGroupStart());
+    NCCL_TRY(ncclSend(src, static_cast<size_t>(bytes), ncclChar, 0, c, s));
+    NCCL_TRY(ncclRecv(dst, static_cast<size_t>(bytes), ncclChar, 0, c, s));
+    ncclResult_t ge = ncclGroupEnd();
+    if (ge != ncclSuccess && ge != ncclInProgress)
+      throw std::runtime_error(std::string("group end: ") +
+                               ncclGetErrorString(ge));
+    wait_comm(c, 30.0, "loopback enqueue");  // same ordering as rccl_send
+    if (hipStreamSynchronize(s) != hipSuccess)
+      throw std::runtime_error("stream sync failed");
+    std::vector<uint8_t> back(static_cast<size_t>(bytes));
+    if (hipMemcpy(back.data(), dst, bytes, hipMemcpyDeviceToHost) != hipSuccess)
+      throw std::runtime_error("D2H failed");
+    if (back != host)
+      throw std::runtime_error("loopback payload mismatch");
+  } catch (...) {
+    cleanup();
+    throw;
+  }
+  cleanup();
 }
 
 void rccl_destroy(RcclLink* l) {

@@ -45,5 +45,6 @@ void rccl_destroy(RcclLink* l);  // abort + free
 // all-reduce.  Validates librccl + the non-blocking poll path without a
 // second GPU; throws on failure.
 void rccl_self_test(int device);
+void rccl_loopback_payload(int device, int64_t bytes);
 
 }  // namespace shamd

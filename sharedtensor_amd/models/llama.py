@@ -70,14 +70,16 @@ def rope_freqs(head_dim, max_seq, theta, device):
 
 
 def apply_rope(x, cos, sin):
-    # x: (B, H, T, D); pairwise rotate [x0, x1] halves interleaved as (even, odd)
+    # x: (B, H, T, D); pairwise rotate [x0, x1] halves interleaved as (even, odd).
+    # Rotation runs in fp32 (cos/sin cache dtype); the result returns to
+    # x's dtype so q/k match v for SDPA outside autocast.
     T = x.shape[2]
     c, s = cos[:T], sin[:T]  # (T, D/2)
     x1, x2 = x[..., 0::2], x[..., 1::2]
     o1 = x1 * c - x2 * s
     o2 = x1 * s + x2 * c
     out = torch.stack((o1, o2), dim=-1).flatten(-2)
-    return out
+    return out.to(x.dtype)
 
 
 class LlamaAttention(nn.Module):
