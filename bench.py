@@ -384,7 +384,9 @@ def run_paramsync(args, rank, world, device):
         provision_up=rank > 0,
         explicit_parent=f"127.0.0.1:{port_base + tree_parent(rank)}" if rank else "",
         listen_port=port_base + rank if world > 1 else 0)
-    sh._start()
+    tj0 = time.perf_counter()
+    sh._start()  # non-masters: join walk + full snapshot state transfer
+    join_s = time.perf_counter() - tj0
     dist = dist_setup(world)
     delta = torch.randn(n, dtype=torch.float32, device=device) * 0.01
 
@@ -427,6 +429,8 @@ def run_paramsync(args, rank, world, device):
             "numel": n, "codec": args.codec,
             "wire_gbps": round(wire / dt / 1e9, 3),
             "staleness_p50": s1["staleness_p50"],
+            # worst rank's join incl. snapshot state transfer (master ~0)
+            "join_s": round(max_over_ranks(dist, join_s if rank else 0.0), 2),
         },
     }
     sh.close()
