@@ -1,9 +1,58 @@
 #include "codec_cpu.h"
 
+#include <atomic>
 #include <cmath>
+#include <cstdlib>
 #include <cstring>
+#include <thread>
+#include <vector>
 
 namespace shamd {
+
+// ------------------------------------------------------------ parallel-for
+static constexpr int64_t PF_CHUNK = int64_t(1) << 22;  // 4M elems, 64-aligned
+
+static int cpu_codec_threads() {
+  static const int t = [] {
+    const char* e = std::getenv("SHTENS_CPU_THREADS");
+    if (e) {
+      int v = std::atoi(e);
+      return v < 1 ? 1 : v;
+    }
+    unsigned hw = std::thread::hardware_concurrency();
+    // cap: several link threads may run codecs concurrently and the engine
+    // shares the host with training; raise via SHTENS_CPU_THREADS for
+    // dedicated 100 GB-scale CPU replicas (BASELINE config 5 rehearsal)
+    return static_cast<int>(hw < 1 ? 1 : (hw > 16 ? 16 : hw));
+  }();
+  return t;
+}
+
+void cpu_pfor(int64_t n, const std::function<void(int64_t, int64_t)>& fn) {
+  if (n <= 0) return;
+  int64_t nchunks = (n + PF_CHUNK - 1) / PF_CHUNK;
+  int nt = cpu_codec_threads();
+  if (nchunks == 1 || nt == 1) {
+    fn(0, n);
+    return;
+  }
+  if (static_cast<int64_t>(nt) > nchunks) nt = static_cast<int>(nchunks);
+  std::atomic<int64_t> next{0};
+  auto worker = [&] {
+    for (;;) {
+      int64_t c = next.fetch_add(1, std::memory_order_relaxed);
+      if (c >= nchunks) return;
+      int64_t lo = c * PF_CHUNK;
+      int64_t hi = lo + PF_CHUNK < n ? lo + PF_CHUNK : n;
+      fn(lo, hi);
+    }
+  };
+  std::vector<std::thread> ths;
+  ths.reserve(nt - 1);
+  for (int i = 1; i < nt; ++i) ths.emplace_back(worker);
+  worker();
+  for (auto& t : ths) t.join();
+}
 
 // ----------------------------------------------------------------- fp8 e4m3
 uint8_t f32_to_e4m3(float xf) {
@@ -60,20 +109,46 @@ float pow2_ceil_f(double x) {
 float cpu_compute_scale(Codec c, const float* delta, int64_t n, int stride) {
   if (n <= 0) return 0.0f;
   if (stride < 1) stride = 1;
+  // per-chunk partials combined IN CHUNK ORDER: the result is bit-identical
+  // regardless of worker count (chunk decomposition depends only on n)
+  int64_t nchunks = (n + PF_CHUNK - 1) / PF_CHUNK;
   if (c == Codec::OneBit) {
+    std::vector<double> part(static_cast<size_t>(nchunks), 0.0);
+    std::vector<int64_t> cnts(static_cast<size_t>(nchunks), 0);
+    cpu_pfor(n, [&](int64_t lo, int64_t hi) {
+      int64_t ci = lo / PF_CHUNK;
+      // strided sampling restarts on the chunk-local grid; with stride=1
+      // (the default) this is the exact sum of squares
+      double ss = 0.0;
+      int64_t cnt = 0;
+      for (int64_t i = lo; i < hi; i += stride, ++cnt) {
+        double d = atomic_load_f32(delta + i);
+        ss += d * d;
+      }
+      part[static_cast<size_t>(ci)] = ss;
+      cnts[static_cast<size_t>(ci)] = cnt;
+    });
     double ss = 0.0;
     int64_t cnt = 0;
-    for (int64_t i = 0; i < n; i += stride, ++cnt) {
-      double d = atomic_load_f32(delta + i);
-      ss += d * d;
+    for (int64_t ci = 0; ci < nchunks; ++ci) {
+      ss += part[static_cast<size_t>(ci)];
+      cnt += cnts[static_cast<size_t>(ci)];
     }
     return pow2_floor_f(std::sqrt(ss / static_cast<double>(cnt)));
   }
+  std::vector<float> partm(static_cast<size_t>(nchunks), 0.0f);
+  cpu_pfor(n, [&](int64_t lo, int64_t hi) {
+    int64_t ci = lo / PF_CHUNK;
+    float mx = 0.0f;
+    for (int64_t i = lo; i < hi; i += stride) {
+      float a = std::fabs(atomic_load_f32(delta + i));
+      if (a > mx) mx = a;
+    }
+    partm[static_cast<size_t>(ci)] = mx;
+  });
   float mx = 0.0f;
-  for (int64_t i = 0; i < n; i += stride) {
-    float a = std::fabs(atomic_load_f32(delta + i));
-    if (a > mx) mx = a;
-  }
+  for (float m : partm)
+    if (m > mx) mx = m;
   if (mx == 0.0f || std::isnan(mx) || std::isinf(mx)) return 0.0f;
   return pow2_ceil_f(mx / (c == Codec::Fp8 ? 448.0 : 7.0));
 }
@@ -83,45 +158,53 @@ void cpu_quantize(Codec c, float* delta, int64_t n, float scale, uint8_t* payloa
   int64_t pb = payload_bytes(c, n);
   std::memset(payload, 0, pb);
   if (scale == 0.0f) return;
+  // parallel over 64-aligned chunks: no packed-payload byte straddles a
+  // chunk, so the |= writes below are worker-exclusive
   switch (c) {
     case Codec::OneBit:
       // residual > 0 -> bit 0, send +scale; else bit 1, send -scale
       // (sharedtensor.c:166-174); debit with an atomic so adds that land
       // between the read and the update are never lost.
-      for (int64_t i = 0; i < n; ++i) {
-        float v = atomic_load_f32(delta + i);
-        float sent;
-        if (v > 0.0f) {
-          sent = scale;
-        } else {
-          payload[i / 8] |= 1u << (i % 8);
-          sent = -scale;
+      cpu_pfor(n, [&](int64_t lo, int64_t hi) {
+        for (int64_t i = lo; i < hi; ++i) {
+          float v = atomic_load_f32(delta + i);
+          float sent;
+          if (v > 0.0f) {
+            sent = scale;
+          } else {
+            payload[i / 8] |= 1u << (i % 8);
+            sent = -scale;
+          }
+          atomic_add_f32(delta + i, -sent);
         }
-        atomic_add_f32(delta + i, -sent);
-      }
+      });
       break;
     case Codec::Fp8: {
       float inv = 1.0f / scale;
-      for (int64_t i = 0; i < n; ++i) {
-        float v = atomic_load_f32(delta + i);
-        float sc = v * inv;
-        sc = sc > 448.f ? 448.f : (sc < -448.f ? -448.f : sc);
-        uint8_t q = f32_to_e4m3(sc);
-        payload[i] = q;
-        atomic_add_f32(delta + i, -e4m3_to_f32(q) * scale);
-      }
+      cpu_pfor(n, [&](int64_t lo, int64_t hi) {
+        for (int64_t i = lo; i < hi; ++i) {
+          float v = atomic_load_f32(delta + i);
+          float sc = v * inv;
+          sc = sc > 448.f ? 448.f : (sc < -448.f ? -448.f : sc);
+          uint8_t q = f32_to_e4m3(sc);
+          payload[i] = q;
+          atomic_add_f32(delta + i, -e4m3_to_f32(q) * scale);
+        }
+      });
       break;
     }
     case Codec::Int4: {
       float inv = 1.0f / scale;
-      for (int64_t i = 0; i < n; ++i) {
-        float v = atomic_load_f32(delta + i);
-        float r = std::nearbyintf(v * inv);  // round-to-nearest-even
-        r = r > 7.f ? 7.f : (r < -7.f ? -7.f : r);
-        int8_t q = static_cast<int8_t>(r);
-        payload[i / 2] |= static_cast<uint8_t>(q & 0xF) << ((i % 2) * 4);
-        atomic_add_f32(delta + i, -static_cast<float>(q) * scale);
-      }
+      cpu_pfor(n, [&](int64_t lo, int64_t hi) {
+        for (int64_t i = lo; i < hi; ++i) {
+          float v = atomic_load_f32(delta + i);
+          float r = std::nearbyintf(v * inv);  // round-to-nearest-even
+          r = r > 7.f ? 7.f : (r < -7.f ? -7.f : r);
+          int8_t q = static_cast<int8_t>(r);
+          payload[i / 2] |= static_cast<uint8_t>(q & 0xF) << ((i % 2) * 4);
+          atomic_add_f32(delta + i, -static_cast<float>(q) * scale);
+        }
+      });
       break;
     }
   }
@@ -131,35 +214,40 @@ void cpu_quantize(Codec c, float* delta, int64_t n, float scale, uint8_t* payloa
 void cpu_apply(Codec c, const uint8_t* payload, int64_t n, float scale,
                float* const* dsts, int ndst) {
   if (scale == 0.0f) return;
-  for (int64_t i = 0; i < n; ++i) {
-    float v = 0.0f;
-    switch (c) {
-      case Codec::OneBit: {
-        int bit = (payload[i / 8] >> (i % 8)) & 1;
-        v = bit ? -scale : scale;  // save_deltas, sharedtensor.c:106-111
-        break;
+  cpu_pfor(n, [&](int64_t lo, int64_t hi) {
+    for (int64_t i = lo; i < hi; ++i) {
+      float v = 0.0f;
+      switch (c) {
+        case Codec::OneBit: {
+          int bit = (payload[i / 8] >> (i % 8)) & 1;
+          v = bit ? -scale : scale;  // save_deltas, sharedtensor.c:106-111
+          break;
+        }
+        case Codec::Fp8:
+          v = e4m3_to_f32(payload[i]) * scale;
+          break;
+        case Codec::Int4: {
+          int8_t q =
+              static_cast<int8_t>((payload[i / 2] >> ((i % 2) * 4)) & 0xF);
+          if (q > 7) q -= 16;
+          v = static_cast<float>(q) * scale;
+          break;
+        }
       }
-      case Codec::Fp8:
-        v = e4m3_to_f32(payload[i]) * scale;
-        break;
-      case Codec::Int4: {
-        int8_t q = static_cast<int8_t>((payload[i / 2] >> ((i % 2) * 4)) & 0xF);
-        if (q > 7) q -= 16;
-        v = static_cast<float>(q) * scale;
-        break;
-      }
+      if (v != 0.0f)
+        for (int d = 0; d < ndst; ++d) atomic_add_f32(dsts[d] + i, v);
     }
-    if (v != 0.0f)
-      for (int d = 0; d < ndst; ++d) atomic_add_f32(dsts[d] + i, v);
-  }
+  });
 }
 
 void cpu_add_scatter(const float* src, int64_t n, float* const* dsts, int ndst) {
-  for (int64_t i = 0; i < n; ++i) {
-    float v = src[i];
-    if (v != 0.0f)
-      for (int d = 0; d < ndst; ++d) atomic_add_f32(dsts[d] + i, v);
-  }
+  cpu_pfor(n, [&](int64_t lo, int64_t hi) {
+    for (int64_t i = lo; i < hi; ++i) {
+      float v = src[i];
+      if (v != 0.0f)
+        for (int d = 0; d < ndst; ++d) atomic_add_f32(dsts[d] + i, v);
+    }
+  });
 }
 
 }  // namespace shamd

@@ -5,6 +5,7 @@
 #pragma once
 
 #include <cstdint>
+#include <functional>
 
 #include "common.h"
 
@@ -33,6 +34,14 @@ inline float atomic_load_f32(const float* p) {
   __builtin_memcpy(&f, &u, 4);
   return f;
 }
+
+// Chunked parallel-for over [0, n): fixed 4M-element chunks (64-aligned so
+// packed-payload bytes are never shared across workers) executed by up to
+// SHTENS_CPU_THREADS workers (default min(hw, 16)).  The chunk decomposition
+// depends only on n, so chunk-local results are machine-independent; the
+// reference's whole engine is one thread per link (sharedtensor.c) — this is
+// where the CPU engine beats it by an order of magnitude on big tensors.
+void cpu_pfor(int64_t n, const std::function<void(int64_t, int64_t)>& fn);
 
 // OCP fp8 e4m3fn conversions (shared semantics with torch.float8_e4m3fn;
 // round-to-nearest-even, caller pre-clamps to +-448).

@@ -485,8 +485,10 @@ void Engine::handshake_as_child(int fd, bool rejoin) {
         }
       } else {
         tmp_host.resize(static_cast<size_t>(n_));
-        for (int64_t i = 0; i < n_; ++i)
-          tmp_host[i] = atomic_load_f32(fdelta(up.delta) + i);
+        cpu_pfor(n_, [&](int64_t lo, int64_t hi) {
+          for (int64_t i = lo; i < hi; ++i)
+            tmp_host[i] = atomic_load_f32(fdelta(up.delta) + i);
+        });
         captured = true;
       }
       if (captured) {
@@ -506,14 +508,16 @@ void Engine::handshake_as_child(int fd, bool rejoin) {
         (void)hipFree(tmp_dev);
         tmp_dev = nullptr;
       } else {
-        for (int64_t i = 0; i < n_; ++i) {
-          float v = tmp_host[i];
-          if (v == 0.0f) continue;
-          atomic_add_f32(values_ + i, v);
-          atomic_add_f32(fdelta(up.delta) + i, v);
-          if (fwd0) atomic_add_f32(fdelta(fwd0) + i, v);
-          if (fwd1) atomic_add_f32(fdelta(fwd1) + i, v);
-        }
+        cpu_pfor(n_, [&](int64_t lo, int64_t hi) {
+          for (int64_t i = lo; i < hi; ++i) {
+            float v = tmp_host[i];
+            if (v == 0.0f) continue;
+            atomic_add_f32(values_ + i, v);
+            atomic_add_f32(fdelta(up.delta) + i, v);
+            if (fwd0) atomic_add_f32(fdelta(fwd0) + i, v);
+            if (fwd1) atomic_add_f32(fdelta(fwd1) + i, v);
+          }
+        });
       }
     };
     if (captured) {
@@ -692,12 +696,14 @@ void Engine::reconnect_loop() try {
                               cfg_.delta_bf16, nullptr);
               HIP_TRY(hipStreamSynchronize(nullptr));
             } else {
-              for (int64_t i = 0; i < n_; ++i) {
-                float v = atomic_load_f32(values_ + i);
-                if (v == 0.0f) continue;
-                if (fwd0) atomic_add_f32(fdelta(fwd0) + i, v);
-                if (fwd1) atomic_add_f32(fdelta(fwd1) + i, v);
-              }
+              cpu_pfor(n_, [&](int64_t lo, int64_t hi) {
+                for (int64_t i = lo; i < hi; ++i) {
+                  float v = atomic_load_f32(values_ + i);
+                  if (v == 0.0f) continue;
+                  if (fwd0) atomic_add_f32(fdelta(fwd0) + i, v);
+                  if (fwd1) atomic_add_f32(fdelta(fwd1) + i, v);
+                }
+              });
             }
           }
           break;
@@ -928,10 +934,12 @@ void Engine::rebuild_slot_invariant(Link& lk) {
                     cfg_.delta_bf16, nullptr);
     HIP_TRY(hipStreamSynchronize(nullptr));
   } else {
-    for (int64_t i = 0; i < n_; ++i) {
-      float v = atomic_load_f32(values_ + i);
-      if (v != 0.0f) atomic_add_f32(fdelta(lk.delta) + i, v);
-    }
+    cpu_pfor(n_, [&](int64_t lo, int64_t hi) {
+      for (int64_t i = lo; i < hi; ++i) {
+        float v = atomic_load_f32(values_ + i);
+        if (v != 0.0f) atomic_add_f32(fdelta(lk.delta) + i, v);
+      }
+    });
   }
 }
 
@@ -1060,7 +1068,7 @@ void Engine::send_snapshot(Link& lk) {
     const char* e = std::getenv("SHTENS_TEST_SNAPSHOT_DELAY_MS");
     return e ? std::atof(e) / 1e3 : 0.0;
   }();
-  int64_t chunk_bytes = gpu() ? (SA_ + P_) : (1 << 20);
+  int64_t chunk_bytes = gpu() ? (SA_ + P_) : (1 << 26);
   if (test_delay_s > 0.0) chunk_bytes = std::min<int64_t>(chunk_bytes, 1 << 14);
   const int64_t chunk_elems = std::max<int64_t>(chunk_bytes / 4, 1);
   std::vector<uint8_t> tmp;
@@ -1082,14 +1090,19 @@ void Engine::send_snapshot(Link& lk) {
                       cfg_.delta_bf16, lk.s_send);
       HIP_TRY(hipStreamSynchronize(lk.s_send));
     } else {
-      for (int64_t i = 0; i < ce; ++i)
-        reinterpret_cast<float*>(tmp.data())[i] = atomic_load_f32(values_ + off + i);
+      cpu_pfor(ce, [&](int64_t lo, int64_t hi) {
+        for (int64_t i = lo; i < hi; ++i)
+          reinterpret_cast<float*>(tmp.data())[i] =
+              atomic_load_f32(values_ + off + i);
+      });
       if (!io_write(lk.fd, tmp.data(), ce * 4))
         throw std::runtime_error("tcp write failed");
       const float* snap = reinterpret_cast<const float*>(tmp.data());
-      for (int64_t i = 0; i < ce; ++i)
-        if (snap[i] != 0.0f)
-          atomic_add_f32(fdelta(lk.delta) + off + i, -snap[i]);
+      cpu_pfor(ce, [&](int64_t lo, int64_t hi) {
+        for (int64_t i = lo; i < hi; ++i)
+          if (snap[i] != 0.0f)
+            atomic_add_f32(fdelta(lk.delta) + off + i, -snap[i]);
+      });
     }
     lk.bytes_sent += ce * 4;
     if (test_delay_s > 0.0)
@@ -1099,7 +1112,7 @@ void Engine::send_snapshot(Link& lk) {
 
 void Engine::recv_snapshot(int fd) {
   Link& up = links_[LK_UP];
-  const int64_t chunk_bytes = gpu() ? (SA_ + P_) : (1 << 20);
+  const int64_t chunk_bytes = gpu() ? (SA_ + P_) : (1 << 26);
   const int64_t chunk_elems = std::max<int64_t>(chunk_bytes / 4, 1);
   std::vector<uint8_t> tmp;
   if (!gpu()) tmp.resize(static_cast<size_t>(chunk_elems) * 4);
@@ -1125,13 +1138,15 @@ void Engine::recv_snapshot(int fd) {
       if (!io_read(fd, tmp.data(), ce * 4))
         throw std::runtime_error("snapshot read failed");
       const float* snap = reinterpret_cast<const float*>(tmp.data());
-      for (int64_t i = 0; i < ce; ++i) {
-        float v = snap[i];
-        if (v == 0.0f) continue;
-        atomic_add_f32(values_ + off + i, v);
-        if (fwd[0]) atomic_add_f32(fdelta(fwd[0]) + off + i, v);
-        if (fwd[1]) atomic_add_f32(fdelta(fwd[1]) + off + i, v);
-      }
+      cpu_pfor(ce, [&](int64_t lo, int64_t hi) {
+        for (int64_t i = lo; i < hi; ++i) {
+          float v = snap[i];
+          if (v == 0.0f) continue;
+          atomic_add_f32(values_ + off + i, v);
+          if (fwd[0]) atomic_add_f32(fdelta(fwd[0]) + off + i, v);
+          if (fwd[1]) atomic_add_f32(fdelta(fwd[1]) + off + i, v);
+        }
+      });
     }
     up.bytes_recv += ce * 4;
   }
@@ -1483,7 +1498,9 @@ void Engine::copy_to(uintptr_t dst, int64_t n, uintptr_t stream) {
                            reinterpret_cast<hipStream_t>(stream)));
   } else {
     float* d = reinterpret_cast<float*>(dst);
-    for (int64_t i = 0; i < n_; ++i) d[i] = atomic_load_f32(values_ + i);
+    cpu_pfor(n_, [&](int64_t lo, int64_t hi) {
+      for (int64_t i = lo; i < hi; ++i) d[i] = atomic_load_f32(values_ + i);
+    });
   }
 }
 
@@ -1501,15 +1518,17 @@ void Engine::fused_sgd(uintptr_t mom, uintptr_t grad, double lr,
   } else {
     float* m = reinterpret_cast<float*>(mom);
     const float* g = reinterpret_cast<const float*>(grad);
-    for (int64_t i = 0; i < n_; ++i) {
-      float mm = static_cast<float>(momentum) * m[i] + g[i];
-      m[i] = mm;
-      float u = static_cast<float>(-lr) * mm;
-      if (u == 0.0f) continue;
-      atomic_add_f32(values_ + i, u);
-      for (int k = 0; k < 3; ++k)
-        if (d[k]) atomic_add_f32(fdelta(d[k]) + i, u);
-    }
+    cpu_pfor(n_, [&](int64_t lo, int64_t hi) {
+      for (int64_t i = lo; i < hi; ++i) {
+        float mm = static_cast<float>(momentum) * m[i] + g[i];
+        m[i] = mm;
+        float u = static_cast<float>(-lr) * mm;
+        if (u == 0.0f) continue;
+        atomic_add_f32(values_ + i, u);
+        for (int k = 0; k < 3; ++k)
+          if (d[k]) atomic_add_f32(fdelta(d[k]) + i, u);
+      }
+    });
   }
   notify_all_dirty();
 }
