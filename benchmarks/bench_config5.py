@@ -1,0 +1,158 @@
+#!/usr/bin/env python3
+"""BASELINE config 5 at verbatim scale on one box: a 100 GB shared tensor
+(25e9 fp32 elements) with int4 wire compression and bf16 residuals on the
+GPU master, plus a CPU-replica child (host RAM) joining over loopback TCP.
+
+Two 100 GB GPU replicas cannot share one 288 GB device (2 x (values 100 +
+bf16 delta 50 + int4 staging 25) GB), so the single-box rehearsal pairs the
+GPU engine with a CPU child — the GPU side runs the real config-5 data path
+(hipMalloc'd 100 GB replica, bf16 residuals, int4 CDNA4 quantize/apply,
+snapshot debit kernels); steady-state GB/s is transport/CPU-peer bound and
+is labeled as such.  On the 8-GPU node each GPU holds its own 100 GB replica
+(values+delta+staging = 175 GB < 288 GB) and peers over real xGMI.
+
+Measures:
+  * snapshot-join wall time at 100 GB (round-1 unknown)
+  * steady-state logical/wire GB/s + staleness p50 over a timed window
+"""
+import argparse
+import json
+import multiprocessing as mp
+import os
+import sys
+import time
+
+os.environ.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
+
+import torch  # noqa: E402
+
+sys.path.insert(0, __file__.rsplit("/", 2)[0])
+
+
+def child_main(port, n, q, threads):
+    os.environ["SHTENS_CPU_THREADS"] = str(threads)
+    from sharedtensor_amd.engine import SharedFlat
+    try:
+        sh = SharedFlat("127.0.0.1", port, [n], device="cpu", codec="int4",
+                        provision_up=True, expected_children=0,
+                        join_timeout_s=1800)
+        t0 = time.perf_counter()
+        sh._start()  # join walk + 100 GB snapshot stream + apply
+        join_s = time.perf_counter() - t0
+        q.put(("joined", join_s))
+        # steady state: the recv loop applies full-size int4 rounds in the
+        # background; idle here until the parent signals stop
+        for _ in range(1800):
+            if os.path.exists("/tmp/shtens_cfg5_stop"):
+                break
+            time.sleep(1)
+        s = sh.stats()
+        q.put(("stats", {"rounds_recv": s["rounds_recv"],
+                         "bytes_recv": s["bytes_recv"],
+                         "staleness_p50": s["staleness_p50"]}))
+        sh.close()
+        q.put(("closed", None))
+    except Exception as e:
+        q.put(("fail", repr(e)))
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gb", type=float, default=100.0)
+    ap.add_argument("--window", type=float, default=45.0,
+                    help="steady-state measurement window (s)")
+    ap.add_argument("--poke-elems", type=int, default=2_000_000_000,
+                    help="bf16 residual elements refreshed per round "
+                         "(keeps full-size packets flowing)")
+    ap.add_argument("--child-threads", type=int, default=64)
+    args = ap.parse_args()
+    n = int(args.gb * 1e9 / 4)
+    port = 56231
+    if os.path.exists("/tmp/shtens_cfg5_stop"):
+        os.unlink("/tmp/shtens_cfg5_stop")
+
+    from sharedtensor_amd.engine import SharedFlat
+    torch.cuda.set_device(0)
+    t0 = time.perf_counter()
+    master = SharedFlat("127.0.0.1", port, [n], device="cuda:0", codec="int4",
+                        delta_dtype=torch.bfloat16, lagged_scale=True,
+                        expected_children=1, provision_up=False)
+    master._start()
+    alloc_s = time.perf_counter() - t0
+    print(f"[cfg5] master up: n={n} ({n*4/1e9:.0f} GB), alloc+start "
+          f"{alloc_s:.1f}s", file=sys.stderr, flush=True)
+    # seed non-zero state in place (no extra 100 GB temp): the snapshot must
+    # stream real data and the child must pay the real apply cost
+    g = torch.Generator(device="cuda").manual_seed(1)
+    step = 1 << 28
+    for off in range(0, n, step):
+        m = min(step, n - off)
+        master.values[off:off + m].normal_(0.0, 0.01, generator=g)
+    torch.cuda.synchronize()
+    print("[cfg5] seeded values", file=sys.stderr, flush=True)
+
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    p = ctx.Process(target=child_main, args=(port, n, q, args.child_threads))
+    p.start()
+    tag, join_s = q.get(timeout=1800)
+    if tag != "joined":
+        print(json.dumps({"error": join_s}))
+        p.kill()
+        return 1
+    print(f"[cfg5] child joined in {join_s:.1f}s "
+          f"({n*4/1e9/join_s:.2f} GB/s snapshot stream)",
+          file=sys.stderr, flush=True)
+
+    # steady state: refresh part of the child-link residual each round so
+    # full-size int4 packets keep flowing (bandwidth measurement; replica
+    # consistency is not the point of this mode and is not asserted)
+    delta_buf = master._link_bufs[0][0]  # bf16[n] residual of the child link
+    s0 = master.stats()
+    t1 = time.perf_counter()
+    rounds_done = s0["rounds_sent"]
+    while time.perf_counter() - t1 < args.window:
+        m = min(args.poke_elems, delta_buf.numel())
+        delta_buf[:m].normal_(0.0, 0.01, generator=g)
+        torch.cuda.synchronize()
+        master.notify()
+        time.sleep(0.5)
+    s1 = master.stats()
+    dt = time.perf_counter() - t1
+    open("/tmp/shtens_cfg5_stop", "w").close()
+    tag, cs = q.get(timeout=120)
+    child_stats = cs if tag == "stats" else {"error": cs}
+    rounds = s1["rounds_sent"] - s0["rounds_sent"]
+    wire = s1["bytes_sent"] - s0["bytes_sent"]
+    out = {
+        "metric": "config5 100GB int4+bf16 paramsync (GPU master + CPU child, 1 box)",
+        "numel": n,
+        "values_gb": round(n * 4 / 1e9, 1),
+        "codec": "int4",
+        "delta_dtype": "bf16",
+        "snapshot_join_s": round(join_s, 1),
+        "snapshot_stream_gbps": round(n * 4 / 1e9 / join_s, 2),
+        "window_s": round(dt, 1),
+        "rounds_sent": rounds,
+        "logical_gbps": round(rounds * n * 4 / dt / 1e9, 2),
+        "wire_gbps": round(wire / dt / 1e9, 2),
+        "staleness_p50_sent": s1["sent_scale_p50"],
+        "child": child_stats,
+        "note": "steady-state bound by loopback TCP + CPU-peer apply; "
+                "GPU side is the config-5 data path (100 GB HBM replica, "
+                "bf16 residuals, int4 CDNA4 codec)",
+    }
+    try:
+        q.get(timeout=60)  # closed
+    except Exception:
+        pass
+    p.join(timeout=60)
+    if p.is_alive():
+        p.kill()
+    master.close()
+    print(json.dumps(out))
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())
