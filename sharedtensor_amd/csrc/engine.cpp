@@ -1068,42 +1068,92 @@ void Engine::send_snapshot(Link& lk) {
     const char* e = std::getenv("SHTENS_TEST_SNAPSHOT_DELAY_MS");
     return e ? std::atof(e) / 1e3 : 0.0;
   }();
-  int64_t chunk_bytes = gpu() ? (SA_ + P_) : (1 << 26);
-  if (test_delay_s > 0.0) chunk_bytes = std::min<int64_t>(chunk_bytes, 1 << 14);
-  const int64_t chunk_elems = std::max<int64_t>(chunk_bytes / 4, 1);
-  std::vector<uint8_t> tmp;
-  if (!gpu()) tmp.resize(static_cast<size_t>(chunk_elems) * 4);
+  if (gpu()) {
+    // Pipelined GPU path: the fused capture+debit kernel writes the
+    // authoritative sent-bytes into half of send_buf, D2H lands it in the
+    // matching half of send_pin, and the TCP write of chunk k overlaps the
+    // capture+D2H of chunk k+1 (double buffering; for small tensors one
+    // chunk covers everything and this degenerates to the simple path).
+    HIP_TRY(hipSetDevice(cfg_.device));
+    // device staging (send_buf, SA_+P_ bytes) holds both halves
+    int64_t half_bytes = std::min<int64_t>((SA_ + P_) / 2, 256 << 20);
+    if (test_delay_s > 0.0)
+      half_bytes = std::min<int64_t>(half_bytes, 1 << 14);
+    const int64_t sub = std::max<int64_t>(half_bytes / 4, 1);
+    hipEvent_t ev[2];
+    HIP_TRY(hipEventCreateWithFlags(&ev[0], hipEventDisableTiming));
+    HIP_TRY(hipEventCreateWithFlags(&ev[1], hipEventDisableTiming));
+    int64_t prev_off = -1, prev_ce = 0;
+    bool ok = true;
+    std::string err;
+    int64_t nstages = (n_ + sub - 1) / sub;
+    for (int64_t k = 0; k < nstages && ok; ++k) {
+      int64_t off = k * sub;
+      int64_t ce = std::min(sub, n_ - off);
+      int b = static_cast<int>(k & 1);
+      auto* dev = reinterpret_cast<float*>(lk.send_buf) + b * sub;
+      uint8_t* pin = lk.send_pin + b * sub * 4;
+      try {
+        hip_snapshot_capture(values_ + off, doff(lk.delta, off),
+                             cfg_.delta_bf16, dev, ce, lk.s_send);
+        HIP_TRY(hipMemcpyAsync(pin, dev, ce * 4, hipMemcpyDeviceToHost,
+                               lk.s_send));
+        HIP_TRY(hipEventRecord(ev[b], lk.s_send));
+      } catch (const std::exception& e) {
+        ok = false;
+        err = e.what();
+        break;
+      }
+      if (prev_off >= 0) {  // write chunk k-1 while chunk k copies
+        uint8_t* ppin = lk.send_pin + ((k - 1) & 1) * sub * 4;
+        HIP_TRY(hipEventSynchronize(ev[(k - 1) & 1]));
+        if (!io_write(lk.fd, ppin, prev_ce * 4)) {
+          ok = false;
+          err = "tcp write failed";
+          break;
+        }
+        lk.bytes_sent += prev_ce * 4;
+        if (test_delay_s > 0.0)
+          std::this_thread::sleep_for(
+              std::chrono::duration<double>(test_delay_s));
+      }
+      prev_off = off;
+      prev_ce = ce;
+    }
+    if (ok && prev_off >= 0) {
+      uint8_t* ppin = lk.send_pin + ((nstages - 1) & 1) * sub * 4;
+      HIP_TRY(hipEventSynchronize(ev[(nstages - 1) & 1]));
+      if (!io_write(lk.fd, ppin, prev_ce * 4)) {
+        ok = false;
+        err = "tcp write failed";
+      } else {
+        lk.bytes_sent += prev_ce * 4;
+      }
+    }
+    (void)hipStreamSynchronize(lk.s_send);  // drain before buffer reuse
+    (void)hipEventDestroy(ev[0]);
+    (void)hipEventDestroy(ev[1]);
+    if (!ok) throw std::runtime_error(err);
+    return;
+  }
+  const int64_t chunk_elems =
+      std::max<int64_t>((test_delay_s > 0.0 ? (1 << 14) : (1 << 26)) / 4, 1);
+  std::vector<uint8_t> tmp(static_cast<size_t>(chunk_elems) * 4);
   for (int64_t off = 0; off < n_; off += chunk_elems) {
     int64_t ce = std::min(chunk_elems, n_ - off);
-    if (gpu()) {
-      HIP_TRY(hipSetDevice(cfg_.device));
-      HIP_TRY(hipMemcpyAsync(lk.send_pin, values_ + off, ce * 4,
-                             hipMemcpyDeviceToHost, lk.s_send));
-      HIP_TRY(hipStreamSynchronize(lk.s_send));
-      if (!io_write(lk.fd, lk.send_pin, ce * 4))
-        throw std::runtime_error("tcp write failed");
-      // debit: delta -= exactly-what-was-sent
-      HIP_TRY(hipMemcpyAsync(lk.send_buf, lk.send_pin, ce * 4,
-                             hipMemcpyHostToDevice, lk.s_send));
-      hip_add_scatter(reinterpret_cast<float*>(lk.send_buf), ce, -1.0f,
-                      nullptr, doff(lk.delta, off), nullptr, nullptr,
-                      cfg_.delta_bf16, lk.s_send);
-      HIP_TRY(hipStreamSynchronize(lk.s_send));
-    } else {
-      cpu_pfor(ce, [&](int64_t lo, int64_t hi) {
-        for (int64_t i = lo; i < hi; ++i)
-          reinterpret_cast<float*>(tmp.data())[i] =
-              atomic_load_f32(values_ + off + i);
-      });
-      if (!io_write(lk.fd, tmp.data(), ce * 4))
-        throw std::runtime_error("tcp write failed");
-      const float* snap = reinterpret_cast<const float*>(tmp.data());
-      cpu_pfor(ce, [&](int64_t lo, int64_t hi) {
-        for (int64_t i = lo; i < hi; ++i)
-          if (snap[i] != 0.0f)
-            atomic_add_f32(fdelta(lk.delta) + off + i, -snap[i]);
-      });
-    }
+    cpu_pfor(ce, [&](int64_t lo, int64_t hi) {
+      for (int64_t i = lo; i < hi; ++i)
+        reinterpret_cast<float*>(tmp.data())[i] =
+            atomic_load_f32(values_ + off + i);
+    });
+    if (!io_write(lk.fd, tmp.data(), ce * 4))
+      throw std::runtime_error("tcp write failed");
+    const float* snap = reinterpret_cast<const float*>(tmp.data());
+    cpu_pfor(ce, [&](int64_t lo, int64_t hi) {
+      for (int64_t i = lo; i < hi; ++i)
+        if (snap[i] != 0.0f)
+          atomic_add_f32(fdelta(lk.delta) + off + i, -snap[i]);
+    });
     lk.bytes_sent += ce * 4;
     if (test_delay_s > 0.0)
       std::this_thread::sleep_for(std::chrono::duration<double>(test_delay_s));

@@ -65,6 +65,11 @@ void hip_add_scatter(const float* src, int64_t n, float alpha, float* values,
                      void* d1, void* d2, void* d3, bool delta_bf16,
                      hipStream_t s);
 
+// Fused join-snapshot capture: out := values (atomic 32-bit loads) and
+// delta -= out in one pass; `out` is the authoritative sent-bytes buffer.
+void hip_snapshot_capture(const float* values, void* delta, bool delta_bf16,
+                          float* out, int64_t n, hipStream_t s);
+
 // {values, d1, d2} += src_delta (delta-typed source; rejoin reconciliation).
 void hip_add_delta_scatter(const void* src_delta, bool delta_bf16, int64_t n,
                            float* values, void* d1, void* d2, hipStream_t s);

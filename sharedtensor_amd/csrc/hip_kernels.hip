@@ -445,6 +445,25 @@ __global__ void k_add_delta_scatter(const typename DP::T* __restrict__ src,
   }
 }
 
+// Fused snapshot capture + debit (join fast path, engine.cpp
+// send_snapshot): out[i] := values[i] and delta[i] -= out[i] in ONE pass.
+// The captured buffer IS the authoritative exactly-what-was-sent bytes, so
+// the old capture -> D2H -> TCP -> H2D -> debit round trip collapses to
+// capture+debit -> D2H -> TCP, and chunks pipeline against the socket.
+template <typename DP>
+__global__ void k_snapshot_capture(const float* __restrict__ values,
+                                   typename DP::T* delta,
+                                   float* __restrict__ out, int64_t n) {
+  int64_t gstride = static_cast<int64_t>(gridDim.x) * blockDim.x;
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+       i += gstride) {
+    float v = __hip_atomic_load(values + i, __ATOMIC_RELAXED,
+                                __HIP_MEMORY_SCOPE_AGENT);
+    out[i] = v;
+    if (v != 0.0f) DP::atomic_add(delta, i, -v);
+  }
+}
+
 template <typename DP>
 __global__ void k_fused_sgd(float* __restrict__ mom,
                             const float* __restrict__ grad, float lr,
@@ -614,6 +633,19 @@ void hip_add_scatter(const float* src, int64_t n, float alpha, float* values,
                        dim3(BLOCK), 0, s, src, n, alpha, values,
                        static_cast<float*>(d1), static_cast<float*>(d2),
                        static_cast<float*>(d3));
+  HIP_CHECK(hipGetLastError());
+}
+
+void hip_snapshot_capture(const float* values, void* delta, bool delta_bf16,
+                          float* out, int64_t n, hipStream_t s) {
+  if (delta_bf16)
+    hipLaunchKernelGGL((k_snapshot_capture<DeltaBF16>), dim3(grid_for(n)),
+                       dim3(BLOCK), 0, s, values,
+                       static_cast<uint16_t*>(delta), out, n);
+  else
+    hipLaunchKernelGGL((k_snapshot_capture<DeltaF32>), dim3(grid_for(n)),
+                       dim3(BLOCK), 0, s, values, static_cast<float*>(delta),
+                       out, n);
   HIP_CHECK(hipGetLastError());
 }
 
