@@ -1079,7 +1079,11 @@ void Engine::send_snapshot(Link& lk) {
     int64_t half_bytes = std::min<int64_t>((SA_ + P_) / 2, 256 << 20);
     if (test_delay_s > 0.0)
       half_bytes = std::min<int64_t>(half_bytes, 1 << 14);
-    const int64_t sub = std::max<int64_t>(half_bytes / 4, 1);
+    // chunk offsets must stay EVEN (bf16 residual debits use packed-pair
+    // atomics on the (even, odd) uint32 — an odd element offset faults) and
+    // prefer 64-alignment; SA_+P_ >= 16 for any codec so sub >= 2 fits
+    int64_t sub = std::max<int64_t>(half_bytes / 4, 2);
+    sub = sub >= 64 ? (sub & ~int64_t(63)) : (sub & ~int64_t(1));
     hipEvent_t ev[2];
     HIP_TRY(hipEventCreateWithFlags(&ev[0], hipEventDisableTiming));
     HIP_TRY(hipEventCreateWithFlags(&ev[1], hipEventDisableTiming));
@@ -1163,7 +1167,11 @@ void Engine::send_snapshot(Link& lk) {
 void Engine::recv_snapshot(int fd) {
   Link& up = links_[LK_UP];
   const int64_t chunk_bytes = gpu() ? (SA_ + P_) : (1 << 26);
-  const int64_t chunk_elems = std::max<int64_t>(chunk_bytes / 4, 1);
+  // even (prefer 64-aligned) chunk offsets: bf16 packed-pair atomics in the
+  // forward-scatter need pair-aligned bases (see send_snapshot)
+  int64_t chunk_elems = std::max<int64_t>(chunk_bytes / 4, 2);
+  chunk_elems = chunk_elems >= 64 ? (chunk_elems & ~int64_t(63))
+                                  : (chunk_elems & ~int64_t(1));
   std::vector<uint8_t> tmp;
   if (!gpu()) tmp.resize(static_cast<size_t>(chunk_elems) * 4);
   void* fwd[2] = {
