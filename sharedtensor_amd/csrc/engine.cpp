@@ -1015,10 +1015,10 @@ bool Engine::rccl_wanted(const Hello& h) const {
   // teardown) on a single leased GPU.  Same-device pairs are declined by
   // default — on one device the TCP loopback path is both correct and has
   // no xGMI to win back.
-  static const bool force_same_dev = [] {
-    const char* e = std::getenv("SHTENS_RCCL_FORCE_SAME_DEVICE");
-    return e && e[0] == '1';
-  }();
+  // read per call, NOT latched in a static: tests toggle this env var
+  // within one process lifetime
+  const char* fsd = std::getenv("SHTENS_RCCL_FORCE_SAME_DEVICE");
+  const bool force_same_dev = fsd && fsd[0] == '1';
   return gpu() && cfg_.use_rccl && rccl_failures_.load() < 2 &&
          (h.flags & HELLO_WANT_RCCL) && h.hostid == hostid_ && h.device >= 0 &&
          (h.device != cfg_.device || force_same_dev);
@@ -1063,11 +1063,10 @@ void Engine::ctrl_loop(Link& lk) {
 
 void Engine::send_snapshot(Link& lk) {
   // test-only join-storm knob: per-chunk throttle (and smaller chunks) to
-  // make a snapshot artificially slow; never set in production paths
-  static const double test_delay_s = [] {
-    const char* e = std::getenv("SHTENS_TEST_SNAPSHOT_DELAY_MS");
-    return e ? std::atof(e) / 1e3 : 0.0;
-  }();
+  // make a snapshot artificially slow; never set in production paths.
+  // Read per call (not a latched static): tests toggle it mid-process.
+  const char* tde = std::getenv("SHTENS_TEST_SNAPSHOT_DELAY_MS");
+  const double test_delay_s = tde ? std::atof(tde) / 1e3 : 0.0;
   if (gpu()) {
     // Pipelined GPU path: the fused capture+debit kernel writes the
     // authoritative sent-bytes into half of send_buf, D2H lands it in the
