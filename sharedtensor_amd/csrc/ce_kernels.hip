@@ -9,6 +9,7 @@
 //   bwd: one pass writing bf16 dlogits = g * (softmax - onehot)
 #include <hip/hip_runtime.h>
 
+#include <atomic>
 #include <stdexcept>
 #include <string>
 
@@ -201,34 +202,38 @@ void hip_ce_fwd(const void* logits, const int32_t* targets, float* loss,
                 float* row_m, float* row_lse, int64_t R, int64_t V,
                 hipStream_t s) {
   // dynamic LDS for the staged variant: one u32 per bf16 pair (+1 pair of
-  // slack for an unaligned head); opt in to >64 KB once per process
+  // slack for an unaligned head).  The >48 KB opt-in attribute is a CUDA-ism
+  // hip may reject — a failed set must not leave a sticky error, and a
+  // rejected LDS launch demotes this process to the two-pass kernel.
   const size_t lds_bytes = (static_cast<size_t>(V) / 2 + 2) * 4;
   static const size_t lds_max = [] {
     int dev = 0;
     (void)hipGetDevice(&dev);
     int v = 0;
     if (hipDeviceGetAttribute(&v, hipDeviceAttributeMaxSharedMemoryPerBlock,
-                              dev) != hipSuccess)
+                              dev) != hipSuccess) {
+      (void)hipGetLastError();
       return size_t(0);
-    size_t cap = static_cast<size_t>(v);
-    if (hipFuncSetAttribute(
-            reinterpret_cast<const void*>(&k_ce_fwd_lds),
-            hipFuncAttributeMaxDynamicSharedMemorySize,
-            static_cast<int>(cap)) != hipSuccess)
-      return size_t(0);
-    return cap;
+    }
+    (void)hipFuncSetAttribute(reinterpret_cast<const void*>(&k_ce_fwd_lds),
+                              hipFuncAttributeMaxDynamicSharedMemorySize, v);
+    (void)hipGetLastError();  // clear any sticky rejection
+    return static_cast<size_t>(v);
   }();
-  if (lds_bytes <= lds_max) {
+  static std::atomic<bool> lds_rejected{false};
+  if (!lds_rejected.load(std::memory_order_relaxed) && lds_bytes <= lds_max) {
     hipLaunchKernelGGL(k_ce_fwd_lds, dim3(static_cast<uint32_t>(R)),
                        dim3(CE_BLOCK), lds_bytes, s,
                        static_cast<const uint16_t*>(logits), targets, loss,
                        row_m, row_lse, V);
-  } else {
-    hipLaunchKernelGGL(k_ce_fwd, dim3(static_cast<uint32_t>(R)),
-                       dim3(CE_BLOCK), 0, s,
-                       static_cast<const uint16_t*>(logits), targets, loss,
-                       row_m, row_lse, V);
+    hipError_t e = hipGetLastError();
+    if (e == hipSuccess) return;
+    lds_rejected.store(true, std::memory_order_relaxed);
   }
+  hipLaunchKernelGGL(k_ce_fwd, dim3(static_cast<uint32_t>(R)),
+                     dim3(CE_BLOCK), 0, s,
+                     static_cast<const uint16_t*>(logits), targets, loss,
+                     row_m, row_lse, V);
   HIP_CHECK_CE(hipGetLastError());
 }
 
