@@ -9,7 +9,6 @@
 //   bwd: one pass writing bf16 dlogits = g * (softmax - onehot)
 #include <hip/hip_runtime.h>
 
-#include <atomic>
 #include <stdexcept>
 #include <string>
 
@@ -110,58 +109,6 @@ __global__ void k_ce_fwd(const uint16_t* __restrict__ logits,
   }
 }
 
-// LDS-staged forward: the whole bf16 row (GPT-2: 50257 elems ~ 100 KB) is
-// staged into LDS during the max pass, so the exp-sum pass never touches
-// global memory again — one HBM read per element instead of the two-pass
-// variant's L2-or-HBM re-read (160 KiB LDS per CU on gfx950 makes a whole
-// vocab row resident; occupancy drops to 1 block/CU but the kernel is
-// bandwidth- not latency-bound).
-__global__ void k_ce_fwd_lds(const uint16_t* __restrict__ logits,
-                             const int32_t* __restrict__ targets,
-                             float* __restrict__ loss,
-                             float* __restrict__ row_m,
-                             float* __restrict__ row_lse, int64_t V) {
-  extern __shared__ uint32_t row_lds[];
-  __shared__ float lds4[4];
-  const int64_t r = blockIdx.x;
-  const uint16_t* xr = logits + r * V;
-  const int head = static_cast<int>(reinterpret_cast<uintptr_t>(xr) & 3) ? 1 : 0;
-  const int64_t npairs = (V - head) >> 1;
-  const bool tail = ((V - head) & 1) != 0;
-  const uint32_t* xp = reinterpret_cast<const uint32_t*>(xr + head);
-
-  float m = -INFINITY;
-  for (int64_t p = threadIdx.x; p < npairs; p += CE_BLOCK) {
-    uint32_t u = xp[p];
-    row_lds[p] = u;
-    m = fmaxf(m, fmaxf(ce_pair_lo(u), ce_pair_hi(u)));
-  }
-  if (threadIdx.x == 0 && head) m = fmaxf(m, ce_bf16_to_f32(xr[0]));
-  if (threadIdx.x == 1 && tail) m = fmaxf(m, ce_bf16_to_f32(xr[V - 1]));
-  for (int w = 32; w > 0; w >>= 1) m = fmaxf(m, __shfl_down(m, w, 64));
-  int wave = threadIdx.x >> 6;
-  if ((threadIdx.x & 63) == 0) lds4[wave] = m;
-  __syncthreads();
-  m = fmaxf(fmaxf(lds4[0], lds4[1]), fmaxf(lds4[2], lds4[3]));
-  __syncthreads();
-
-  float s = 0.f;
-  for (int64_t p = threadIdx.x; p < npairs; p += CE_BLOCK) {
-    uint32_t u = row_lds[p];
-    s += __expf(ce_pair_lo(u) - m) + __expf(ce_pair_hi(u) - m);
-  }
-  if (threadIdx.x == 0 && head) s += __expf(ce_bf16_to_f32(xr[0]) - m);
-  if (threadIdx.x == 1 && tail) s += __expf(ce_bf16_to_f32(xr[V - 1]) - m);
-  float tot = block_sum(s, lds4);
-  if (threadIdx.x == 0) {
-    float lse = __logf(tot) + m;
-    float xt = ce_bf16_to_f32(xr[targets[r]]);
-    loss[r] = lse - xt;
-    row_m[r] = m;
-    row_lse[r] = lse;
-  }
-}
-
 __global__ void k_ce_bwd(const uint16_t* __restrict__ logits,
                          const int32_t* __restrict__ targets,
                          const float* __restrict__ row_lse,
@@ -201,35 +148,11 @@ __global__ void k_ce_bwd(const uint16_t* __restrict__ logits,
 void hip_ce_fwd(const void* logits, const int32_t* targets, float* loss,
                 float* row_m, float* row_lse, int64_t R, int64_t V,
                 hipStream_t s) {
-  // dynamic LDS for the staged variant: one u32 per bf16 pair (+1 pair of
-  // slack for an unaligned head).  The >48 KB opt-in attribute is a CUDA-ism
-  // hip may reject — a failed set must not leave a sticky error, and a
-  // rejected LDS launch demotes this process to the two-pass kernel.
-  const size_t lds_bytes = (static_cast<size_t>(V) / 2 + 2) * 4;
-  static const size_t lds_max = [] {
-    int dev = 0;
-    (void)hipGetDevice(&dev);
-    int v = 0;
-    if (hipDeviceGetAttribute(&v, hipDeviceAttributeMaxSharedMemoryPerBlock,
-                              dev) != hipSuccess) {
-      (void)hipGetLastError();
-      return size_t(0);
-    }
-    (void)hipFuncSetAttribute(reinterpret_cast<const void*>(&k_ce_fwd_lds),
-                              hipFuncAttributeMaxDynamicSharedMemorySize, v);
-    (void)hipGetLastError();  // clear any sticky rejection
-    return static_cast<size_t>(v);
-  }();
-  static std::atomic<bool> lds_rejected{false};
-  if (!lds_rejected.load(std::memory_order_relaxed) && lds_bytes <= lds_max) {
-    hipLaunchKernelGGL(k_ce_fwd_lds, dim3(static_cast<uint32_t>(R)),
-                       dim3(CE_BLOCK), lds_bytes, s,
-                       static_cast<const uint16_t*>(logits), targets, loss,
-                       row_m, row_lse, V);
-    hipError_t e = hipGetLastError();
-    if (e == hipSuccess) return;
-    lds_rejected.store(true, std::memory_order_relaxed);
-  }
+  // NOTE (measured, profiles/r02): an LDS-staged single-read variant
+  // (whole 100 KB row resident in the 160 KiB LDS) ran 4x SLOWER than this
+  // two-pass kernel -- at 100 KB/block only one block fits per CU and four
+  // waves cannot hide the global-load latency; the second pass's L2 re-read
+  // is cheaper than the lost occupancy.
   hipLaunchKernelGGL(k_ce_fwd, dim3(static_cast<uint32_t>(R)),
                      dim3(CE_BLOCK), 0, s,
                      static_cast<const uint16_t*>(logits), targets, loss,
