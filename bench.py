@@ -211,6 +211,7 @@ def run_train(args, rank, world, device):
         model, host="127.0.0.1",
         port_base=int(os.environ.get("SHTENS_PORT_BASE", 52000)),
         rank=rank, world=world, lr=args.lr, momentum=0.9,
+        optimizer=args.opt, weight_decay=0.01 if args.opt == "adamw" else 0.0,
         amp_dtype=torch.bfloat16 if device.type == "cuda" else None,
         param_dtype=torch.bfloat16 if use_bf16_params else torch.float32,
         codec=args.codec, use_rccl=not args.no_rccl,
@@ -472,6 +473,8 @@ def main():
     ap.add_argument("--seq", type=int, default=1024)
     ap.add_argument("--lr", type=float, default=0.01)
     ap.add_argument("--codec", choices=["1bit", "fp8", "int4"], default="1bit")
+    ap.add_argument("--opt", choices=["sgd", "adamw"], default="sgd",
+                    help="fused optimizer feeding the shared tensor")
     ap.add_argument("--model", choices=["small", "tiny", "llama1b", "llama8b"], default="small")
     ap.add_argument("--mode", choices=["train", "paramsync", "table"], default="train")
     ap.add_argument("--numel", type=int, default=268_435_456)  # 1 GB fp32

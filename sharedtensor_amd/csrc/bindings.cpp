@@ -178,6 +178,8 @@ PYBIND11_MODULE(_core, m) {
            py::call_guard<py::gil_scoped_release>())
       .def("fused_sgd_bf16", &Engine::fused_sgd_bf16,
            py::call_guard<py::gil_scoped_release>())
+      .def("fused_adamw", &Engine::fused_adamw,
+           py::call_guard<py::gil_scoped_release>())
       .def("notify_dirty", &Engine::notify_dirty)
       .def("close", &Engine::close, py::call_guard<py::gil_scoped_release>())
       .def("is_master", &Engine::is_master)

@@ -1608,6 +1608,57 @@ void Engine::fused_sgd_bf16(uintptr_t mom, uintptr_t grad_bf16,
   notify_all_dirty();
 }
 
+void Engine::fused_adamw(uintptr_t mom, uintptr_t vel, uintptr_t grad,
+                         bool grad_bf16, uintptr_t shadow, double lr,
+                         double beta1, double beta2, double eps, double wd,
+                         int64_t step, uintptr_t stream) {
+  if (step < 1) throw std::runtime_error("fused_adamw: step must be >= 1");
+  if (grad_bf16 && !gpu())
+    throw std::runtime_error("bf16 grads are a GPU-only path");
+  std::shared_lock<std::shared_mutex> ug(user_m_);
+  void* d[3];
+  for (int i = 0; i < 3; ++i)
+    d[i] = links_[i].provisioned ? links_[i].delta : nullptr;
+  const float b1 = static_cast<float>(beta1), b2 = static_cast<float>(beta2);
+  const float inv_bc1 =
+      1.0f / (1.0f - std::pow(b1, static_cast<float>(step)));
+  const float inv_bc2 =
+      1.0f / (1.0f - std::pow(b2, static_cast<float>(step)));
+  if (gpu()) {
+    hip_fused_adamw(reinterpret_cast<float*>(mom),
+                    reinterpret_cast<float*>(vel),
+                    reinterpret_cast<const void*>(grad), grad_bf16,
+                    reinterpret_cast<uint16_t*>(shadow),
+                    static_cast<float>(lr), b1, b2, static_cast<float>(eps),
+                    static_cast<float>(wd), inv_bc1, inv_bc2, n_, values_,
+                    d[0], d[1], d[2], cfg_.delta_bf16,
+                    reinterpret_cast<hipStream_t>(stream));
+  } else {
+    float* m = reinterpret_cast<float*>(mom);
+    float* v = reinterpret_cast<float*>(vel);
+    const float* g = reinterpret_cast<const float*>(grad);
+    const float lrf = static_cast<float>(lr), epsf = static_cast<float>(eps);
+    const float wdf = static_cast<float>(wd);
+    cpu_pfor(n_, [&](int64_t lo, int64_t hi) {
+      for (int64_t i = lo; i < hi; ++i) {
+        float gi = g[i];
+        float mi = b1 * m[i] + (1.f - b1) * gi;
+        m[i] = mi;
+        float vi = b2 * v[i] + (1.f - b2) * gi * gi;
+        v[i] = vi;
+        float w = atomic_load_f32(values_ + i);
+        float u = -lrf * (mi * inv_bc1 / (std::sqrt(vi * inv_bc2) + epsf) +
+                          wdf * w);
+        if (u == 0.0f) continue;
+        atomic_add_f32(values_ + i, u);
+        for (int k = 0; k < 3; ++k)
+          if (d[k]) atomic_add_f32(fdelta(d[k]) + i, u);
+      }
+    });
+  }
+  notify_all_dirty();
+}
+
 std::vector<LinkStatsSnap> Engine::link_stats() {
   std::vector<LinkStatsSnap> out;
   for (auto& lk : links_) {

@@ -82,6 +82,10 @@ class Engine {
                  uintptr_t stream);
   void fused_sgd_bf16(uintptr_t mom, uintptr_t grad_bf16, uintptr_t shadow_bf16,
                       double lr, double momentum, uintptr_t stream);
+  // shadow==0: fp32 grads, no shadow refresh (grad_bf16 must be false)
+  void fused_adamw(uintptr_t mom, uintptr_t vel, uintptr_t grad, bool grad_bf16,
+                   uintptr_t shadow, double lr, double beta1, double beta2,
+                   double eps, double wd, int64_t step, uintptr_t stream);
   void notify_dirty();  // wake senders after out-of-band delta writes
   void close();
 

@@ -88,6 +88,16 @@ void hip_fused_sgd_bf16(float* mom, const uint16_t* grad, uint16_t* shadow,
                         void* d1, void* d2, void* d3, bool delta_bf16,
                         hipStream_t s);
 
+// Fused AdamW feeding the shared tensor (torch.optim.AdamW semantics:
+// decoupled weight decay on the pre-update weight): m/v fp32 state, grads
+// fp32 or bf16, optional bf16 shadow refresh, update applied to values and
+// staged into the link deltas — one HBM pass.  inv_bc* = 1/(1-beta*^t).
+void hip_fused_adamw(float* mom, float* vel, const void* grad, bool grad_bf16,
+                     uint16_t* shadow, float lr, float beta1, float beta2,
+                     float eps, float wd, float inv_bc1, float inv_bc2,
+                     int64_t n, float* values, void* d1, void* d2, void* d3,
+                     bool delta_bf16, hipStream_t s);
+
 // Fused bf16 LayerNorm for the training path (ln_kernels.hip): fwd saves
 // fp32 mean/rstd; bwd = dx pass + register-accumulated dgamma/dbeta pass.
 void hip_ln_fwd(const void* x, const void* w, const void* b, void* y,

@@ -28,6 +28,7 @@
 #include <hip/hip_bf16.h>
 #include <hip/hip_runtime.h>
 
+#include <type_traits>
 #include <stdexcept>
 #include <string>
 
@@ -661,6 +662,67 @@ void hip_add_delta_scatter(const void* src_delta, bool delta_bf16, int64_t n,
                        dim3(BLOCK), 0, s, static_cast<const float*>(src_delta),
                        n, values, static_cast<float*>(d1),
                        static_cast<float*>(d2));
+  HIP_CHECK(hipGetLastError());
+}
+
+// Fused AdamW feeding the shared tensor — one HBM pass over {m, v, grad,
+// values, link deltas}: decoupled weight decay against the pre-update
+// master weight (torch.optim.AdamW semantics), fp32 m/v state, update
+// applied to the replica with atomicAdd (concurrent gossip folds in) and
+// staged into every link residual.  inv_bc1/2 = 1/(1-beta^t) host-side.
+__device__ __forceinline__ float gval(const float* g, int64_t i) { return g[i]; }
+__device__ __forceinline__ float gval(const uint16_t* g, int64_t i) {
+  return bf16_to_f32(g[i]);
+}
+
+template <typename DP, typename GT>
+__global__ void k_fused_adamw(float* __restrict__ mom, float* __restrict__ vel,
+                              const GT* __restrict__ grad,
+                              uint16_t* __restrict__ shadow, float lr,
+                              float beta1, float beta2, float eps, float wd,
+                              float inv_bc1, float inv_bc2, int64_t n,
+                              float* __restrict__ values, typename DP::T* d1,
+                              typename DP::T* d2, typename DP::T* d3) {
+  int64_t gstride = static_cast<int64_t>(gridDim.x) * blockDim.x;
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+       i += gstride) {
+    float g = gval(grad, i);
+    float m = beta1 * mom[i] + (1.f - beta1) * g;
+    mom[i] = m;
+    float v = beta2 * vel[i] + (1.f - beta2) * g * g;
+    vel[i] = v;
+    float w = __hip_atomic_load(values + i, __ATOMIC_RELAXED,
+                                __HIP_MEMORY_SCOPE_AGENT);
+    float u = -lr * (m * inv_bc1 / (sqrtf(v * inv_bc2) + eps) + wd * w);
+    float old = atomicAdd(values + i, u);
+    if (shadow) shadow[i] = f32_to_bf16(old + u);
+    if (d1) DP::atomic_add(d1, i, u);
+    if (d2) DP::atomic_add(d2, i, u);
+    if (d3) DP::atomic_add(d3, i, u);
+  }
+}
+
+void hip_fused_adamw(float* mom, float* vel, const void* grad, bool grad_bf16,
+                     uint16_t* shadow, float lr, float beta1, float beta2,
+                     float eps, float wd, float inv_bc1, float inv_bc2,
+                     int64_t n, float* values, void* d1, void* d2, void* d3,
+                     bool delta_bf16, hipStream_t s) {
+  auto launch = [&](auto dp_tag, auto g_ptr) {
+    using DP = decltype(dp_tag);
+    hipLaunchKernelGGL((k_fused_adamw<DP, std::remove_pointer_t<decltype(g_ptr)>>),
+                       dim3(grid_for(n)), dim3(BLOCK), 0, s, mom, vel, g_ptr,
+                       shadow, lr, beta1, beta2, eps, wd, inv_bc1, inv_bc2, n,
+                       values, static_cast<typename DP::T*>(d1),
+                       static_cast<typename DP::T*>(d2),
+                       static_cast<typename DP::T*>(d3));
+  };
+  if (delta_bf16) {
+    if (grad_bf16) launch(DeltaBF16{}, static_cast<const uint16_t*>(grad));
+    else launch(DeltaBF16{}, static_cast<const float*>(grad));
+  } else {
+    if (grad_bf16) launch(DeltaF32{}, static_cast<const uint16_t*>(grad));
+    else launch(DeltaF32{}, static_cast<const float*>(grad));
+  }
   HIP_CHECK(hipGetLastError());
 }
 

@@ -183,6 +183,43 @@ class _SharedBase:
                                  shadow_bf16.data_ptr(), float(lr),
                                  float(momentum), self._stream())
 
+    def fused_adamw_step(self, mom: torch.Tensor, vel: torch.Tensor,
+                         grad: torch.Tensor, step: int, lr: float,
+                         betas=(0.9, 0.999), eps: float = 1e-8,
+                         weight_decay: float = 0.0):
+        """torch.optim.AdamW-semantics update feeding the shared tensor:
+        fp32 m/v, decoupled weight decay on the pre-update master weight,
+        update applied to the replica AND staged into every link delta in
+        one HBM pass (HIP kernel k_fused_adamw)."""
+        if not (mom.numel() == vel.numel() == grad.numel() == self.n):
+            raise ValueError("size mismatch")
+        if grad.dtype != torch.float32:
+            raise TypeError("fused_adamw_step takes fp32 grads "
+                            "(use fused_adamw_bf16_step)")
+        self._eng.fused_adamw(mom.data_ptr(), vel.data_ptr(), grad.data_ptr(),
+                              False, 0, float(lr), float(betas[0]),
+                              float(betas[1]), float(eps),
+                              float(weight_decay), int(step), self._stream())
+
+    def fused_adamw_bf16_step(self, mom: torch.Tensor, vel: torch.Tensor,
+                              grad_bf16: torch.Tensor,
+                              shadow_bf16: torch.Tensor, step: int, lr: float,
+                              betas=(0.9, 0.999), eps: float = 1e-8,
+                              weight_decay: float = 0.0):
+        """Mixed-precision fused AdamW: bf16 grads in, fp32 master updated,
+        bf16 shadow params refreshed (folding concurrent gossip), link
+        deltas staged — one HBM pass."""
+        if grad_bf16.dtype != torch.bfloat16 or shadow_bf16.dtype != torch.bfloat16:
+            raise TypeError("grad/shadow must be bfloat16")
+        if not (mom.numel() == vel.numel() == grad_bf16.numel()
+                == shadow_bf16.numel() == self.n):
+            raise ValueError("size mismatch")
+        self._eng.fused_adamw(mom.data_ptr(), vel.data_ptr(),
+                              grad_bf16.data_ptr(), True,
+                              shadow_bf16.data_ptr(), float(lr),
+                              float(betas[0]), float(betas[1]), float(eps),
+                              float(weight_decay), int(step), self._stream())
+
     # -- observability -----------------------------------------------------
     def stats(self) -> dict:
         links = self._eng.link_stats()

@@ -126,6 +126,39 @@ class AsyncSGD:
         self.shared.grad_flat.zero_()
 
 
+class AsyncAdamW:
+    """AdamW whose update feeds the shared tensor through the fused kernel
+    (torch.optim.AdamW semantics: decoupled weight decay on the pre-update
+    master weight) — the optimizer step IS the addFromTensor."""
+
+    def __init__(self, shared: FlatParamShared, lr: float = 3e-4,
+                 betas=(0.9, 0.999), eps: float = 1e-8,
+                 weight_decay: float = 0.01):
+        self.shared = shared
+        self.lr = lr
+        self.betas = betas
+        self.eps = eps
+        self.weight_decay = weight_decay
+        self.step_count = 0
+        self.vel_flat = torch.zeros(shared.n, dtype=torch.float32,
+                                    device=shared.device)
+
+    def step(self):
+        self.step_count += 1
+        sh = self.shared
+        if sh.shadow is not None:
+            sh.fused_adamw_bf16_step(sh.mom_flat, self.vel_flat, sh.grad_flat,
+                                     sh.shadow, self.step_count, self.lr,
+                                     self.betas, self.eps, self.weight_decay)
+        else:
+            sh.fused_adamw_step(sh.mom_flat, self.vel_flat, sh.grad_flat,
+                                self.step_count, self.lr, self.betas,
+                                self.eps, self.weight_decay)
+
+    def zero_grad(self, set_to_none: bool = False):
+        self.shared.grad_flat.zero_()
+
+
 class AsyncDPTrainer:
     """One rank's training loop against the shared parameter tensor."""
 
@@ -133,7 +166,9 @@ class AsyncDPTrainer:
                  port_base: Optional[int] = None, rank: Optional[int] = None,
                  world: Optional[int] = None, lr: float = 0.1,
                  momentum: float = 0.9, amp_dtype: Optional[torch.dtype] = torch.bfloat16,
-                 param_dtype: Optional[torch.dtype] = None, **engine_kw):
+                 param_dtype: Optional[torch.dtype] = None,
+                 optimizer: str = "sgd", betas=(0.9, 0.999),
+                 eps: float = 1e-8, weight_decay: float = 0.0, **engine_kw):
         rank = int(os.environ.get("RANK", 0)) if rank is None else rank
         world = int(os.environ.get("WORLD_SIZE", 1)) if world is None else world
         if port_base is None:
@@ -143,7 +178,13 @@ class AsyncDPTrainer:
             param_dtype = torch.float32
         self.shared = FlatParamShared(model, host, port_base, rank, world,
                                       param_dtype=param_dtype, **engine_kw)
-        self.opt = AsyncSGD(self.shared, lr=lr, momentum=momentum)
+        if optimizer == "adamw":
+            self.opt = AsyncAdamW(self.shared, lr=lr, betas=betas, eps=eps,
+                                  weight_decay=weight_decay)
+        elif optimizer == "sgd":
+            self.opt = AsyncSGD(self.shared, lr=lr, momentum=momentum)
+        else:
+            raise ValueError(f"unknown optimizer {optimizer!r}")
         self.amp_dtype = amp_dtype
         self.rank, self.world = rank, world
         self.device = self.shared.device
