@@ -299,6 +299,14 @@ def run_train(args, rank, world, device):
         for k in ("paramsync_wire_gbps", "paramsync_logical_gbps",
                   "sync_rounds_per_s"):
             result["config"][k] = None
+    else:
+        # rank 0's per-link view (root of the tree): rounds/s per link and
+        # whether the data plane upgraded to RCCL/xGMI
+        result["config"]["links_rank0"] = [
+            {"peer": l1["peer"], "rccl": l1["rccl"],
+             "rounds_sent_per_s": round((l1["rounds_sent"] - l0["rounds_sent"]) / dt, 1),
+             "rounds_recv_per_s": round((l1["rounds_recv"] - l0["rounds_recv"]) / dt, 1)}
+            for l0, l1 in zip(s0["links"], s1["links"]) if l1["active"]]
     trainer.close()
     if dist is not None:
         dist.barrier()
