@@ -99,6 +99,28 @@ def main():
         out[f"torch_rms{C2}_bwd_ms"] = round(t(lambda: torch.autograd.grad(
             yt3, (xt3, wt3), dyr, retain_graph=True)), 3)
 
+    # ---- SwiGLU (llama-1B MLP shape R=8192, F=8192) ----
+    Rs, Fs = 8192, 8192
+    sx1 = torch.randn(Rs, Fs, device="cuda").to(torch.bfloat16)
+    sx3 = torch.randn(Rs, Fs, device="cuda").to(torch.bfloat16)
+    sdy = torch.randn(Rs, Fs, device="cuda").to(torch.bfloat16)
+    sy = torch.empty_like(sx1)
+    sdx1 = torch.empty_like(sx1)
+    sdx3 = torch.empty_like(sx3)
+    ns = sx1.numel()
+    out["swiglu_fwd_ms"] = round(t(lambda: _core.swiglu_fwd(
+        sx1.data_ptr(), sx3.data_ptr(), sy.data_ptr(), ns, s)), 3)
+    out["swiglu_bwd_ms"] = round(t(lambda: _core.swiglu_bwd(
+        sdy.data_ptr(), sx1.data_ptr(), sx3.data_ptr(), sdx1.data_ptr(),
+        sdx3.data_ptr(), ns, s)), 3)
+    sa = sx1.clone().requires_grad_(True)
+    sb = sx3.clone().requires_grad_(True)
+    out["torch_swiglu_fwd_ms"] = round(t(
+        lambda: torch.nn.functional.silu(sa) * sb), 3)
+    syt = torch.nn.functional.silu(sa) * sb
+    out["torch_swiglu_bwd_ms"] = round(t(lambda: torch.autograd.grad(
+        syt, (sa, sb), sdy, retain_graph=True)), 3)
+
     # ---- CE (R=65536, V=50257) ----
     R, V = 65536, 50257
     logits = (torch.randn(R, V, device="cuda") * 2).to(torch.bfloat16)

@@ -29,6 +29,7 @@ SOURCES = [
     "ln_kernels.hip",
     "ce_kernels.hip",
     "gelu_kernels.hip",
+    "swiglu_kernels.hip",
     "bindings.cpp",
 ]
 

@@ -276,6 +276,23 @@ PYBIND11_MODULE(_core, m) {
                R, C, static_cast<float>(eps),
                reinterpret_cast<hipStream_t>(stream));
   });
+  m.def("swiglu_fwd", [](uintptr_t x1, uintptr_t x3, uintptr_t y, int64_t n,
+                         uintptr_t stream) {
+    hip_swiglu_fwd(reinterpret_cast<const void*>(x1),
+                   reinterpret_cast<const void*>(x3),
+                   reinterpret_cast<void*>(y), n,
+                   reinterpret_cast<hipStream_t>(stream));
+  });
+  m.def("swiglu_bwd", [](uintptr_t dy, uintptr_t x1, uintptr_t x3,
+                         uintptr_t dx1, uintptr_t dx3, int64_t n,
+                         uintptr_t stream) {
+    hip_swiglu_bwd(reinterpret_cast<const void*>(dy),
+                   reinterpret_cast<const void*>(x1),
+                   reinterpret_cast<const void*>(x3),
+                   reinterpret_cast<void*>(dx1),
+                   reinterpret_cast<void*>(dx3), n,
+                   reinterpret_cast<hipStream_t>(stream));
+  });
   m.def("rms_fwd", [](uintptr_t x, uintptr_t w, uintptr_t y, uintptr_t rstd,
                       int64_t R, int C, double eps, uintptr_t stream) {
     hip_rms_fwd(reinterpret_cast<const void*>(x),

@@ -125,6 +125,13 @@ void hip_ce_bwd(const void* logits, const int32_t* targets,
                 const float* row_lse, void* dlogits, const float* gscale_dev,
                 float inv_r, int64_t R, int64_t V, hipStream_t s);
 
+// Fused bf16 SwiGLU (swiglu_kernels.hip): y = silu(x1)*x3 one kernel each
+// way; n must be a multiple of 8, buffers 16-byte aligned (uint4 loads).
+void hip_swiglu_fwd(const void* x1, const void* x3, void* y, int64_t n,
+                    hipStream_t s);
+void hip_swiglu_bwd(const void* dy, const void* x1, const void* x3, void* dx1,
+                    void* dx3, int64_t n, hipStream_t s);
+
 // Fused bf16 GELU-tanh (gelu_kernels.hip); n must be even, buffers
 // 4-byte aligned (pair loads).
 void hip_gelu_fwd(const void* x, void* y, int64_t n, hipStream_t s);

@@ -124,7 +124,15 @@ class LlamaMLP(nn.Module):
         self.w2 = nn.Linear(cfg.ffn_dim, cfg.dim, bias=False)  # down
 
     def forward(self, x):
-        return self.w2(F.silu(self.w1(x)) * self.w3(x))
+        x1 = self.w1(x)
+        x3 = self.w3(x)
+        if x1.is_cuda and x1.dtype == torch.bfloat16:
+            # one fused CDNA4 kernel instead of separate silu + mul
+            # (csrc/swiglu_kernels.hip)
+            from ..ops import fused_swiglu
+            if fused_swiglu.can_use(x1, x3):
+                return self.w2(fused_swiglu.fused_swiglu(x1, x3))
+        return self.w2(F.silu(x1) * x3)
 
 
 class LlamaBlock(nn.Module):
