@@ -35,4 +35,4 @@ def fused_gelu(x: torch.Tensor) -> torch.Tensor:
 
 
 def can_use(x: torch.Tensor) -> bool:
-    return x.is_cuda and x.dtype == torch.bfloat16 and x.numel() % 2 == 0
+    return x.is_cuda and x.dtype == torch.bfloat16 and x.numel() % 8 == 0
