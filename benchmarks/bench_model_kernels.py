@@ -99,6 +99,15 @@ def main():
         out[f"torch_rms{C2}_bwd_ms"] = round(t(lambda: torch.autograd.grad(
             yt3, (xt3, wt3), dyr, retain_graph=True)), 3)
 
+    # ---- bias-grad column sum (R=65536, C=768 and 3072) ----
+    for Cc in (768, 3072):
+        Rc = 65536
+        xc = torch.randn(Rc, Cc, device="cuda").to(torch.bfloat16)
+        oc = torch.zeros(Cc, dtype=torch.float32, device="cuda")
+        out[f"colsum{Cc}_ms"] = round(t(lambda: _core.colsum_bf16(
+            xc.data_ptr(), oc.data_ptr(), Rc, Cc, s)), 3)
+        out[f"torch_colsum{Cc}_ms"] = round(t(lambda: xc.sum(0)), 3)
+
     # ---- SwiGLU (llama-1B MLP shape R=8192, F=8192) ----
     Rs, Fs = 8192, 8192
     sx1 = torch.randn(Rs, Fs, device="cuda").to(torch.bfloat16)

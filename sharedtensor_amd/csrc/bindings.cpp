@@ -276,6 +276,12 @@ PYBIND11_MODULE(_core, m) {
                R, C, static_cast<float>(eps),
                reinterpret_cast<hipStream_t>(stream));
   });
+  m.def("colsum_bf16", [](uintptr_t x, uintptr_t out, int64_t R, int C,
+                          uintptr_t stream) {
+    hip_colsum_bf16(reinterpret_cast<const void*>(x),
+                    reinterpret_cast<float*>(out), R, C,
+                    reinterpret_cast<hipStream_t>(stream));
+  });
   m.def("swiglu_fwd", [](uintptr_t x1, uintptr_t x3, uintptr_t y, int64_t n,
                          uintptr_t stream) {
     hip_swiglu_fwd(reinterpret_cast<const void*>(x1),

@@ -107,6 +107,11 @@ void hip_ln_bwd(const void* dy, const void* x, const void* w,
                 const float* mean, const float* rstd, void* dx, float* dgamma,
                 float* dbeta, int64_t R, int C, hipStream_t s);
 
+// Column sum of (R, C) bf16 into fp32 out[C] (Linear backward bias grad);
+// caller zeroes `out` first.
+void hip_colsum_bf16(const void* x, float* out, int64_t R, int C,
+                     hipStream_t s);
+
 // Fused bf16 RMSNorm for the Llama training path (ln_kernels.hip): fwd
 // saves fp32 rstd; bwd = single-reduction dx pass + register-accumulated
 // dgamma pass.
