@@ -9,7 +9,6 @@ buffers, reclaim/rebuild kernels) and reports a JSON verdict for profiles/.
 """
 import json
 import multiprocessing as mp
-import os
 import sys
 import time
 

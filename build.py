@@ -13,7 +13,6 @@ import argparse
 import hashlib
 import os
 import subprocess
-import sys
 import sysconfig
 
 ROOT = os.path.dirname(os.path.abspath(__file__))

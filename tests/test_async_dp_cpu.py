@@ -1,11 +1,9 @@
 """Async-DP trainer wiring on CPU with 2 processes (explicit tree topology —
 the same code path the multi-GPU bench uses, minus the GPU)."""
 import multiprocessing as mp
-import os
 import socket
 import time
 
-import pytest
 import torch
 
 from sharedtensor_amd.models.gpt2 import GPT2, GPT2Config

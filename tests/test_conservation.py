@@ -5,11 +5,9 @@ Multiple threads hammer add_from_tensor concurrently with live gossip; after
 quiescence both replicas must equal the exact sum of everything added
 (error feedback guarantees nothing is lost, only delayed)."""
 import multiprocessing as mp
-import socket
 import threading
 import time
 
-import pytest
 import torch
 
 import sharedtensor_amd as st

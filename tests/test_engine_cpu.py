@@ -6,8 +6,6 @@ join walk (Y/N redirect), snapshot + converging bootstrap, gossip
 convergence, table mode, clean close (no exit(-1)).
 """
 import multiprocessing as mp
-import os
-import socket
 import time
 
 import pytest

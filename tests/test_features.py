@@ -1,15 +1,13 @@
 """Feature tests: bandwidth limiting, sampled RMS, stats surface, example."""
 import multiprocessing as mp
-import socket
 import subprocess
 import sys
 import time
 
-import pytest
 import torch
 
 import sharedtensor_amd as st
-from sharedtensor_amd.utils import free_port, wait_until
+from sharedtensor_amd.utils import free_port
 
 
 

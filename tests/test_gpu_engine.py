@@ -4,8 +4,6 @@ Exercises the full GPU data path: HIP reduce/quantize on the send stream,
 pinned staging, H2D + fused unpack/scatter on the recv stream.
 """
 import multiprocessing as mp
-import os
-import socket
 import time
 
 import pytest

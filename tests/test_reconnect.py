@@ -3,10 +3,8 @@ acknowledged missing feature — any disconnect exits the process,
 /root/reference/src/sharedtensor.c:62,80,99 and README.md:33)."""
 import multiprocessing as mp
 import os
-import socket
 import time
 
-import pytest
 import torch
 
 import sharedtensor_amd as st

@@ -7,7 +7,7 @@ import time
 import torch
 
 import sharedtensor_amd as st
-from sharedtensor_amd.utils import free_port, wait_until
+from sharedtensor_amd.utils import free_port
 
 
 def test_nan_input_does_not_crash_engine():
