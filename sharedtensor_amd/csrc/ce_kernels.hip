@@ -9,6 +9,7 @@
 //   bwd: one pass writing bf16 dlogits = g * (softmax - onehot)
 #include <hip/hip_runtime.h>
 
+#include <cstdlib>
 #include <stdexcept>
 #include <string>
 
@@ -46,15 +47,18 @@ static __device__ __forceinline__ uint32_t ce_pair_pack(float lo, float hi) {
          (static_cast<uint32_t>(ce_f32_to_bf16(hi)) << 16);
 }
 
-constexpr int CE_BLOCK = 256;  // 4 waves
+constexpr int CE_BLOCK = 256;  // 4 waves (fwd: measured vs 512, see launcher)
 
-// block-wide sum (4-wave block): wave shfl tree + LDS combine
-static __device__ __forceinline__ float block_sum(float v, float* lds4) {
+// block-wide sum (BLK/64-wave block): wave shfl tree + LDS combine
+template <int BLK>
+static __device__ __forceinline__ float block_sum(float v, float* ldsw) {
   for (int w = 32; w > 0; w >>= 1) v += __shfl_down(v, w, 64);
   int wave = threadIdx.x >> 6;
-  if ((threadIdx.x & 63) == 0) lds4[wave] = v;
+  if ((threadIdx.x & 63) == 0) ldsw[wave] = v;
   __syncthreads();
-  float t = lds4[0] + lds4[1] + lds4[2] + lds4[3];
+  float t = 0.f;
+#pragma unroll
+  for (int i = 0; i < BLK / 64; ++i) t += ldsw[i];
   __syncthreads();
   return t;
 }
@@ -63,6 +67,7 @@ static __device__ __forceinline__ float block_sum(float v, float* lds4) {
 // row is processed as [optional head elem | aligned uint32 pairs | optional
 // tail elem] — 4-byte loads halve the load-instruction count that bounds
 // these kernels.
+template <int BLK>
 __global__ void k_ce_fwd(const uint16_t* __restrict__ logits,
                          const int32_t* __restrict__ targets,
                          float* __restrict__ loss, float* __restrict__ row_m,
@@ -70,7 +75,7 @@ __global__ void k_ce_fwd(const uint16_t* __restrict__ logits,
   // two independent passes (max, then sum of exp) — an online single pass
   // has a loop-carried (m, s) dependency per thread and measured 5x slower;
   // the second pass re-reads the row from L2 (rows are ~100 KB)
-  __shared__ float lds4[4];
+  __shared__ float lds4[BLK / 64];
   const int64_t r = blockIdx.x;
   const uint16_t* xr = logits + r * V;
   const int head = static_cast<int>(reinterpret_cast<uintptr_t>(xr) & 3) ? 1 : 0;
@@ -79,7 +84,7 @@ __global__ void k_ce_fwd(const uint16_t* __restrict__ logits,
   const uint32_t* xp = reinterpret_cast<const uint32_t*>(xr + head);
 
   float m = -INFINITY;
-  for (int64_t p = threadIdx.x; p < npairs; p += CE_BLOCK) {
+  for (int64_t p = threadIdx.x; p < npairs; p += BLK) {
     uint32_t u = xp[p];
     m = fmaxf(m, fmaxf(ce_pair_lo(u), ce_pair_hi(u)));
   }
@@ -89,17 +94,20 @@ __global__ void k_ce_fwd(const uint16_t* __restrict__ logits,
   int wave = threadIdx.x >> 6;
   if ((threadIdx.x & 63) == 0) lds4[wave] = m;
   __syncthreads();
-  m = fmaxf(fmaxf(lds4[0], lds4[1]), fmaxf(lds4[2], lds4[3]));
+  float mm = -INFINITY;
+#pragma unroll
+  for (int i = 0; i < BLK / 64; ++i) mm = fmaxf(mm, lds4[i]);
+  m = mm;
   __syncthreads();
 
   float s = 0.f;
-  for (int64_t p = threadIdx.x; p < npairs; p += CE_BLOCK) {
+  for (int64_t p = threadIdx.x; p < npairs; p += BLK) {
     uint32_t u = xp[p];
     s += __expf(ce_pair_lo(u) - m) + __expf(ce_pair_hi(u) - m);
   }
   if (threadIdx.x == 0 && head) s += __expf(ce_bf16_to_f32(xr[0]) - m);
   if (threadIdx.x == 1 && tail) s += __expf(ce_bf16_to_f32(xr[V - 1]) - m);
-  float tot = block_sum(s, lds4);
+  float tot = block_sum<BLK>(s, lds4);
   if (threadIdx.x == 0) {
     float lse = __logf(tot) + m;
     float xt = ce_bf16_to_f32(xr[targets[r]]);
@@ -109,6 +117,7 @@ __global__ void k_ce_fwd(const uint16_t* __restrict__ logits,
   }
 }
 
+template <int BLK>
 __global__ void k_ce_bwd(const uint16_t* __restrict__ logits,
                          const int32_t* __restrict__ targets,
                          const float* __restrict__ row_lse,
@@ -127,7 +136,7 @@ __global__ void k_ce_bwd(const uint16_t* __restrict__ logits,
   const bool tail = ((V - head) & 1) != 0;
   const uint32_t* xp = reinterpret_cast<const uint32_t*>(xr + head);
   uint32_t* dp = reinterpret_cast<uint32_t*>(dr + head);
-  for (int64_t p = threadIdx.x; p < npairs; p += CE_BLOCK) {
+  for (int64_t p = threadIdx.x; p < npairs; p += BLK) {
     uint32_t u = xp[p];
     int64_t v0 = head + 2 * p;
     float g0 = gscale * (__expf(ce_pair_lo(u) - lse) - (v0 == tgt ? 1.f : 0.f));
@@ -153,20 +162,37 @@ void hip_ce_fwd(const void* logits, const int32_t* targets, float* loss,
   // two-pass kernel -- at 100 KB/block only one block fits per CU and four
   // waves cannot hide the global-load latency; the second pass's L2 re-read
   // is cheaper than the lost occupancy.
-  hipLaunchKernelGGL(k_ce_fwd, dim3(static_cast<uint32_t>(R)),
-                     dim3(CE_BLOCK), 0, s,
-                     static_cast<const uint16_t*>(logits), targets, loss,
-                     row_m, row_lse, V);
+  // block-size knob for measurement (SHTENS_CE_BLOCK=512); default from
+  // the recorded microbench
+  const char* e = std::getenv("SHTENS_CE_BLOCK");
+  if (e && std::atoi(e) == 512)
+    hipLaunchKernelGGL(k_ce_fwd<512>, dim3(static_cast<uint32_t>(R)),
+                       dim3(512), 0, s,
+                       static_cast<const uint16_t*>(logits), targets, loss,
+                       row_m, row_lse, V);
+  else
+    hipLaunchKernelGGL(k_ce_fwd<CE_BLOCK>, dim3(static_cast<uint32_t>(R)),
+                       dim3(CE_BLOCK), 0, s,
+                       static_cast<const uint16_t*>(logits), targets, loss,
+                       row_m, row_lse, V);
   HIP_CHECK_CE(hipGetLastError());
 }
 
 void hip_ce_bwd(const void* logits, const int32_t* targets,
                 const float* row_lse, void* dlogits, const float* gscale_dev,
                 float inv_r, int64_t R, int64_t V, hipStream_t s) {
-  hipLaunchKernelGGL(k_ce_bwd, dim3(static_cast<uint32_t>(R)), dim3(CE_BLOCK),
-                     0, s, static_cast<const uint16_t*>(logits), targets,
-                     row_lse, static_cast<uint16_t*>(dlogits), gscale_dev,
-                     inv_r, V);
+  const char* e = std::getenv("SHTENS_CE_BLOCK");
+  if (e && std::atoi(e) == 512)
+    hipLaunchKernelGGL(k_ce_bwd<512>, dim3(static_cast<uint32_t>(R)),
+                       dim3(512), 0, s, static_cast<const uint16_t*>(logits),
+                       targets, row_lse, static_cast<uint16_t*>(dlogits),
+                       gscale_dev, inv_r, V);
+  else
+    hipLaunchKernelGGL(k_ce_bwd<CE_BLOCK>, dim3(static_cast<uint32_t>(R)),
+                       dim3(CE_BLOCK), 0, s,
+                       static_cast<const uint16_t*>(logits), targets,
+                       row_lse, static_cast<uint16_t*>(dlogits), gscale_dev,
+                       inv_r, V);
   HIP_CHECK_CE(hipGetLastError());
 }
 
