@@ -162,8 +162,9 @@ void hip_ce_fwd(const void* logits, const int32_t* targets, float* loss,
   // two-pass kernel -- at 100 KB/block only one block fits per CU and four
   // waves cannot hide the global-load latency; the second pass's L2 re-read
   // is cheaper than the lost occupancy.
-  // block-size knob for measurement (SHTENS_CE_BLOCK=512); default from
-  // the recorded microbench
+  // block-size knob (SHTENS_CE_BLOCK=512); measured r02: 512 is within
+  // noise of 256 (2.63 vs 2.66 ms fwd) — the kernel is bound by the
+  // L2 re-read + exp throughput, not wave count — so 256 stays default
   const char* e = std::getenv("SHTENS_CE_BLOCK");
   if (e && std::atoi(e) == 512)
     hipLaunchKernelGGL(k_ce_fwd<512>, dim3(static_cast<uint32_t>(R)),
