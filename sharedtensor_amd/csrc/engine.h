@@ -112,10 +112,16 @@ class Engine {
   int listen_port_ = 0;
   sockaddr_in listen_addr_{};
   std::mutex slots_m_;  // child-slot assignment vs reconnect demotion
-  // User mutation ops (add_from / fused_sgd* / copy_to) hold this shared;
-  // rejoin reconciliation and failover invariant-restore hold it exclusive
-  // so a concurrent update cannot land once in values and twice via a
-  // captured residual.
+  // User mutation ops (add_from / fused_sgd* / fused_adamw / copy_to) and
+  // packet applies hold this shared; rejoin reconciliation, failover
+  // invariant-restore and slot rebuilds hold it exclusive so a concurrent
+  // update cannot land once in values and twice via a captured residual
+  // (ADVICE round 1, medium x2).  Note on fairness: glibc rwlocks prefer
+  // readers, so a steady storm of shared holders could delay the exclusive
+  // taker — acceptable here because shared sections are short (one kernel
+  // enqueue / one packet apply) and exclusive takers only run on rare
+  // failure-recovery paths where the up-link is already down (no inbound
+  // gossip) and user steps leave gaps between ops.
   std::shared_mutex user_m_;
   std::mutex close_m_;  // serializes concurrent close() calls
   std::thread listen_thread_;
