@@ -56,6 +56,11 @@ def test_self_launch_gpus2():
     # with links up, the param-sync half of the metric is measured
     assert d["config"]["paramsync_wire_gbps"] is not None
     assert d["config"]["sync_rounds_per_s"] >= 0
+    links = d["config"]["links_rank0"]  # root's per-link view
+    assert len(links) >= 1
+    for l in links:
+        for key in ("peer", "rccl", "rounds_sent_per_s", "rounds_recv_per_s"):
+            assert key in l, key
 
 
 def test_self_launch_gpus8_dry():
