@@ -505,11 +505,15 @@ def main():
     ap.add_argument("--device", default="auto")
     args = ap.parse_args()
 
-    if "WORLD_SIZE" not in os.environ and args.gpus > 1:
+    # A torchrun-style context is recognized by RANK (always set per rank);
+    # a stray WORLD_SIZE in the parent environment without RANK must NOT
+    # suppress self-launch (that is the round-1 hang, inverted)
+    if "RANK" not in os.environ and args.gpus > 1:
         sys.exit(self_launch(args))
 
     rank = int(os.environ.get("RANK", 0))
-    world = int(os.environ.get("WORLD_SIZE", args.gpus))
+    world = int(os.environ.get("WORLD_SIZE", args.gpus)) \
+        if "RANK" in os.environ else args.gpus
     local = int(os.environ.get("LOCAL_RANK", rank))
     if args.device == "auto":
         # modulo device count so an N-rank rehearsal also runs on fewer GPUs
