@@ -122,3 +122,79 @@ def test_child_rejoins_restarted_master():
     finally:
         child.join(timeout=30)
     assert child.exitcode == 0
+
+
+def _racing_adder(port, q, go_ev, done_ev):
+    """Joins, converges, then adds continuously through a master crash +
+    rejoin window, reporting exactly how much it added."""
+    try:
+        h = st.create_or_fetch("127.0.0.1", port, torch.zeros(256),
+                               reconnect=True, join_timeout_s=60)
+        out = torch.zeros(256)
+
+        def conv(v, tol=1e-2):
+            h.copy_to_tensor(out)
+            return abs(out[0].item() - v) < tol
+
+        if not wait_until(lambda: conv(5.0), timeout=30):
+            q.put(("fail", f"no initial converge: {out[0]}"))
+            return
+        q.put(("joined", None))
+        go_ev.wait(60)  # master is crashing/restarting NOW; keep adding
+        total = 0.0
+        t0 = time.time()
+        while not done_ev.is_set() and time.time() - t0 < 60:
+            h.add_from_tensor(torch.full((256,), 1e-3))
+            total += 1e-3
+            time.sleep(0.005)
+        # exactly-once through the rejoin: every add made during the outage
+        # and the reconciliation window must survive exactly once
+        target = 5.0 + total
+        ok = wait_until(lambda: conv(target, tol=0.02), timeout=30)
+        q.put(("final", (ok, float(out[0]), target, h.stats()["reconnects"],
+                         h.stats()["last_error"])))
+        time.sleep(2)
+        h.close()
+    except Exception as e:  # pragma: no cover
+        q.put(("fail", repr(e)))
+
+
+def test_rejoin_reconciliation_is_exactly_once_under_adds():
+    """ADVICE round-1 medium: an add racing the rejoin reconciliation used
+    to land once in values AND again via the re-added residual.  With the
+    capture-and-swap fix, value(final) == seed + everything added during
+    the outage window — nothing lost, nothing doubled."""
+    port = free_port()
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    go_ev = ctx.Event()
+    done_ev = ctx.Event()
+    m1 = st.create_or_fetch("127.0.0.1", port, torch.full((256,), 5.0))
+    child = ctx.Process(target=_racing_adder, args=(port, q, go_ev, done_ev))
+    child.start()
+    try:
+        stage, _ = q.get(timeout=60)
+        assert stage == "joined"
+        time.sleep(2.0)  # drain: everything delivered to master #1
+        m1._eng.close()  # crash master #1 (its state dies with it)
+        go_ev.set()      # child starts adding into the dead link
+        time.sleep(1.5)
+        # restart the master with the SAME seed: the child's rejoin must
+        # carry (as unsent residual) every add made since the crash
+        m2 = st.create_or_fetch("127.0.0.1", port, torch.full((256,), 5.0),
+                                join_timeout_s=30)
+        # let the child rejoin (or fail over) and keep adding a while
+        time.sleep(4.0)
+        done_ev.set()
+        stage, res = q.get(timeout=90)
+        assert stage == "final", res
+        ok, got, target, reconnects, err = res
+        assert ok, (f"value {got} != seed+adds {target} "
+                    f"(reconnects={reconnects}, err={err})")
+        m2.close()
+    finally:
+        done_ev.set()
+        child.join(timeout=30)
+        if child.is_alive():
+            child.kill()
+    assert child.exitcode == 0
