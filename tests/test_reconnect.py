@@ -198,3 +198,78 @@ def test_rejoin_reconciliation_is_exactly_once_under_adds():
         if child.is_alive():
             child.kill()
     assert child.exitcode == 0
+
+
+def _short_lived_child(port, q):
+    h = st.create_or_fetch("127.0.0.1", port, torch.zeros(128))
+    out = torch.zeros(128)
+
+    def conv():
+        h.copy_to_tensor(out)
+        return abs(out[0].item() - 3.0) < 1e-2
+    q.put(("ok", None) if wait_until(conv, timeout=30) else ("fail", out[0]))
+    h.add_from_tensor(torch.full((128,), 2.0))
+    time.sleep(2)  # let the delta reach the master
+    os._exit(1)    # die hard: no CLOSE, slot goes L_DEAD with residue
+
+
+def _second_child(port, q):
+    try:
+        h = st.create_or_fetch("127.0.0.1", port, torch.zeros(128))
+        out = torch.zeros(128)
+
+        def conv():
+            h.copy_to_tensor(out)
+            return abs(out[0].item() - 6.0) < 1e-2
+        if not wait_until(conv, timeout=30):
+            q.put(("fail", f"never reached 6.0: {out[0]}"))
+            return
+        time.sleep(3)  # stale slot residue would arrive as garbage NOW
+        h.copy_to_tensor(out)
+        drift = abs(out[0].item() - 6.0)
+        q.put(("ok", drift) if drift < 0.05 else
+              ("fail", f"post-join drift {out[0]} (dead-slot residue?)"))
+        h.close()
+    except Exception as e:  # pragma: no cover
+        q.put(("fail", repr(e)))
+
+
+def test_reclaimed_dead_slot_does_not_corrupt_next_joiner():
+    """Regression (found by test_chaos, round 2): a dead child's slot kept
+    its residual delta (undelivered state minus the snapshot debit); the
+    NEXT joiner claiming the slot received that residue as garbage gossip.
+    rebuild_slot_invariant (slot := values at claim) must keep the new
+    joiner exactly at the master's state."""
+    port = free_port()
+    ctx = mp.get_context("spawn")
+    master = st.create_or_fetch("127.0.0.1", port, torch.full((128,), 3.0))
+    q = ctx.Queue()
+    a = ctx.Process(target=_short_lived_child, args=(port, q))
+    a.start()
+    try:
+        status, info = q.get(timeout=60)
+        assert status == "ok", info
+        a.join(timeout=30)  # child A crashed itself
+        out = torch.zeros(128)
+
+        def master_at(v):
+            master.copy_to_tensor(out)
+            return abs(out[0].item() - v) < 1e-2
+        assert wait_until(lambda: master_at(5.0), timeout=30), out[0]
+        # more updates land in the dead slot's residue before reclaim
+        master.add_from_tensor(torch.ones(128))
+        assert wait_until(lambda: master_at(6.0), timeout=10)
+        b = ctx.Process(target=_second_child, args=(port, q))
+        b.start()
+        try:
+            status, info = q.get(timeout=90)
+            assert status == "ok", info
+        finally:
+            b.join(timeout=30)
+            if b.is_alive():
+                b.kill()
+        assert b.exitcode == 0
+    finally:
+        if a.is_alive():
+            a.kill()
+        master.close()
