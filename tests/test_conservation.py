@@ -5,6 +5,7 @@ Multiple threads hammer add_from_tensor concurrently with live gossip; after
 quiescence both replicas must equal the exact sum of everything added
 (error feedback guarantees nothing is lost, only delayed)."""
 import multiprocessing as mp
+import os
 import threading
 import time
 
@@ -104,3 +105,77 @@ def test_save_restore(tmp_path):
         out = torch.zeros(6, 7)
         h2.copy_to_tensor(out)
         assert torch.equal(out, vals)
+
+
+def _join_during_load_child(port, to_master, to_child):
+    try:
+        h = st.create_or_fetch("127.0.0.1", port, torch.zeros(65536))
+        to_master.put(("joined", None))
+        out = torch.zeros(65536)
+        target = to_child.get(timeout=45)[1]
+        deadline = time.time() + 45
+        while time.time() < deadline:
+            h.copy_to_tensor(out)
+            if abs(out[0].item() - target) < 0.02:
+                to_master.put(("ok", float(out[0].item())))
+                time.sleep(2)
+                h.close()
+                return
+            time.sleep(0.05)
+        to_master.put(("fail", f"child at {out[0].item()} target {target}"))
+    except Exception as e:  # pragma: no cover
+        to_master.put(("fail", repr(e)))
+
+
+def test_snapshot_join_under_concurrent_adds_is_exactly_once():
+    """The snapshot fast-join debits the link delta by exactly the bytes
+    sent; adds racing the (artificially slowed) snapshot stream must land
+    exactly once on both replicas: final == seed + total-added."""
+    os.environ["SHTENS_TEST_SNAPSHOT_DELAY_MS"] = "25"
+    try:
+        port = free_port()
+        master = st.create_or_fetch("127.0.0.1", port,
+                                    torch.full((65536,), 2.0))
+        ctx = mp.get_context("spawn")
+        to_master = ctx.Queue()
+        to_child = ctx.Queue()
+        p = ctx.Process(target=_join_during_load_child,
+                        args=(port, to_master, to_child))
+        p.start()
+        try:
+            total = 0.0
+            # add continuously through the child's walk + slowed snapshot
+            t0 = time.time()
+            joined_at = None
+            while time.time() - t0 < 30:
+                master.add_from_tensor(torch.full((65536,), 1e-3))
+                total += 1e-3
+                try:
+                    msg = to_master.get_nowait()
+                    if msg[0] == "joined":
+                        joined_at = time.time()
+                except Exception:
+                    pass
+                if joined_at and time.time() - joined_at > 1.0:
+                    break  # kept adding well past the join
+                time.sleep(0.002)
+            assert joined_at is not None, "child never joined"
+            target = 2.0 + total
+            to_child.put(("target", target))
+            status, info = to_master.get(timeout=60)
+            assert status == "ok", info
+            out = torch.zeros(65536)
+
+            def conv():
+                master.copy_to_tensor(out)
+                return abs(out[0].item() - target) < 0.02
+            assert wait_until(conv, timeout=30), \
+                f"master {out[0].item()} != {target}"
+        finally:
+            p.join(timeout=30)
+            if p.is_alive():
+                p.kill()
+            master.close()
+        assert p.exitcode == 0
+    finally:
+        del os.environ["SHTENS_TEST_SNAPSHOT_DELAY_MS"]
