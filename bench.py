@@ -6,9 +6,9 @@ Contract (driver): `python bench.py --gpus N --steps K --warmup W` runs one
 rank per GPU.  Two launch paths, both supported:
   * torchrun (`torch.distributed.run --nproc-per-node N`): each rank reads
     RANK / WORLD_SIZE / LOCAL_RANK from the env.
-  * plain `python bench.py --gpus N` with no WORLD_SIZE in the env: this
-    process SELF-LAUNCHES N rank subprocesses (env RANK/LOCAL_RANK/
-    WORLD_SIZE/MASTER_*), waits for them, and forwards rank 0's JSON line.
+  * plain `python bench.py --gpus N` with no RANK in the env: this process
+    SELF-LAUNCHES N rank subprocesses (env RANK/LOCAL_RANK/WORLD_SIZE/
+    MASTER_*), waits for them, and forwards rank 0's JSON line.
 Each rank does W untimed warmup steps, times exactly K steps bracketed by
 barrier + torch.cuda.synchronize on both sides, takes the MAX step time over
 ranks, and rank 0 prints ONE JSON line.
