@@ -233,6 +233,17 @@ def run_train(args, rank, world, device):
         loss = trainer.step(b[:, :-1], b[:, 1:])
         return loss
 
+    if device.type == "cuda":
+        # clock warm: fresh leases start in a low-power state and the DVFS
+        # ramp takes ~1-2 s — longer than a few warmup steps; burn a bounded
+        # 2 s of GEMM before the (untimed) warmup so the timed region runs
+        # at steady clocks
+        wa = torch.randn(4096, 4096, device=device, dtype=torch.bfloat16)
+        t_warm = time.perf_counter()
+        while time.perf_counter() - t_warm < 2.0:
+            wa = wa @ wa
+            wa = wa / wa.norm().clamp_min(1e-6)
+        torch.cuda.synchronize()
     log(f"warmup {args.warmup} steps")
     for i in range(args.warmup):
         one_step(i)
