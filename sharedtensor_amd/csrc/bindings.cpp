@@ -152,6 +152,7 @@ PYBIND11_MODULE(_core, m) {
       .def_readwrite("snapshot_join", &Config::snapshot_join)
       .def_readwrite("use_rccl", &Config::use_rccl)
       .def_readwrite("reconnect", &Config::reconnect)
+      .def_readwrite("preserve_subtree", &Config::preserve_subtree)
       .def_readwrite("keepalive_s", &Config::keepalive_s)
       .def_readwrite("bw_limit", &Config::bw_limit)
       .def_readwrite("min_round_interval_s", &Config::min_round_interval_s)

@@ -49,6 +49,13 @@ struct Config {
   bool snapshot_join = true; // 'Y' is followed by a full state snapshot
   bool use_rccl = true;      // allow RCCL/xGMI upgrade for same-host GPU links
   bool reconnect = false;    // try to rejoin if the up-link dies
+  bool preserve_subtree = false;  // on up-link death keep children attached
+                             // and rejoin with the whole subtree: the
+                             // reconciliation computes a correction delta
+                             // (S + R - V_old) that drains to the children
+                             // as ordinary gossip.  Acyclic by construction:
+                             // the detached subtree is unreachable from the
+                             // root's redirect walk.  Requires snapshot_join.
   double keepalive_s = 1.0;  // idle ping interval (reference: 1s, :161-164)
   double bw_limit = 0.0;     // bytes/sec cap per link, 0 = unlimited
                              // (reference TODO, README.md:31)

@@ -450,7 +450,135 @@ void Engine::handshake_as_child(int fd, bool rejoin) {
       throw;
     }
   }
-  if (rejoin && (ah.flags & ACC_SNAPSHOT)) {
+  if (rejoin && (ah.flags & ACC_SNAPSHOT) && rejoin_preserve_) {
+    // Subtree-preserving reconciliation (cfg.preserve_subtree): the
+    // children stayed ATTACHED and keep gossiping.  They must end up
+    // applying exactly corr = S + R - V_old on top of whatever they had:
+    //   phase 1 (exclusive): R := up.delta (scratch fp32); up.delta := 0;
+    //     corr := -values; values := 0.  Child residuals untouched — their
+    //     U_c (what the child still lacks vs OUR values) stays exact.
+    //   phase 2 (lock-free): snapshot chunks add into values AND corr,
+    //     never into child residuals; concurrent user updates u land in
+    //     values + every provisioned residual (including the children's)
+    //     and never touch corr, so corr stays exactly S - V_old.
+    //   phase 3 (exclusive): values += R; up.delta += R; corr += R;
+    //     child.delta += corr for every ACTIVE child.
+    //   End: values = S + R + u; each live child's residual = U_c + u +
+    //     (S + R - V_old) == values - child_view, which drains as ordinary
+    //     gossip — the child never disconnects and never re-snapshots.
+    // On a phase-2 failure, phase 3 still runs with the partial S: the
+    // state is then self-consistent (a truncated snapshot) and the retry
+    // re-captures from it.
+    float* corr = gpu() ? sub_corr_dev_ : sub_corr_host_.data();
+    float* tmpR = nullptr;
+    std::vector<float> tmpR_host;
+    bool captured = false;
+    {
+      std::unique_lock<std::shared_mutex> ug(user_m_);
+      if (gpu()) {
+        HIP_TRY(hipSetDevice(cfg_.device));
+        HIP_TRY(hipDeviceSynchronize());
+        if (hipMalloc(&tmpR, n_ * 4) == hipSuccess) {
+          HIP_TRY(hipMemsetAsync(tmpR, 0, n_ * 4, up.s_recv));
+          hip_add_delta_scatter(up.delta, cfg_.delta_bf16, n_, tmpR, nullptr,
+                                nullptr, up.s_recv);
+          HIP_TRY(hipMemsetAsync(corr, 0, n_ * 4, up.s_recv));
+          hip_add_scatter(values_, n_, -1.0f, corr, nullptr, nullptr,
+                          nullptr, false, up.s_recv);
+          HIP_TRY(hipStreamSynchronize(up.s_recv));
+          captured = true;
+        } else {
+          (void)hipGetLastError();
+          tmpR = nullptr;
+        }
+      } else {
+        tmpR_host.resize(static_cast<size_t>(n_));
+        cpu_pfor(n_, [&](int64_t lo, int64_t hi) {
+          for (int64_t i = lo; i < hi; ++i) {
+            tmpR_host[i] = atomic_load_f32(fdelta(up.delta) + i);
+            corr[i] = -atomic_load_f32(values_ + i);
+          }
+        });
+        captured = true;
+      }
+      if (captured) {
+        zero_delta(up.delta);
+        zero_buf(values_, n_);
+      }
+    }
+    auto phase3 = [&] {
+      std::unique_lock<std::shared_mutex> ug(user_m_);
+      if (gpu()) {
+        HIP_TRY(hipDeviceSynchronize());
+        hip_add_scatter(tmpR, n_, 1.0f, values_, up.delta, nullptr, nullptr,
+                        cfg_.delta_bf16, up.s_recv);
+        hip_add_scatter(tmpR, n_, 1.0f, corr, nullptr, nullptr, nullptr,
+                        false, up.s_recv);
+        for (int i : {LK_LEFT, LK_RIGHT}) {
+          Link& c = links_[i];
+          if (c.state.load() == L_ACTIVE)
+            hip_add_scatter(corr, n_, 1.0f, nullptr, c.delta, nullptr,
+                            nullptr, cfg_.delta_bf16, up.s_recv);
+        }
+        HIP_TRY(hipStreamSynchronize(up.s_recv));
+        (void)hipFree(tmpR);
+        tmpR = nullptr;
+      } else {
+        float* cl = links_[LK_LEFT].state.load() == L_ACTIVE
+                        ? fdelta(links_[LK_LEFT].delta) : nullptr;
+        float* cr = links_[LK_RIGHT].state.load() == L_ACTIVE
+                        ? fdelta(links_[LK_RIGHT].delta) : nullptr;
+        cpu_pfor(n_, [&](int64_t lo, int64_t hi) {
+          for (int64_t i = lo; i < hi; ++i) {
+            float r = tmpR_host[i];
+            if (r != 0.0f) {
+              atomic_add_f32(values_ + i, r);
+              atomic_add_f32(fdelta(up.delta) + i, r);
+              corr[i] += r;
+            }
+            float cv = corr[i];
+            if (cv != 0.0f) {
+              if (cl) atomic_add_f32(cl + i, cv);
+              if (cr) atomic_add_f32(cr + i, cv);
+            }
+          }
+        });
+      }
+      notify_all_dirty();  // children have fresh residual to drain
+    };
+    if (captured) {
+      try {
+        recv_snapshot(fd, corr);
+      } catch (...) {
+        phase3();  // consistent truncated-snapshot state for the retry
+        throw;
+      }
+      phase3();
+    } else {
+      // no scratch for R: single exclusive section, R stays in up.delta
+      std::unique_lock<std::shared_mutex> ug(user_m_);
+      HIP_TRY(hipDeviceSynchronize());
+      HIP_TRY(hipMemsetAsync(corr, 0, n_ * 4, up.s_recv));
+      hip_add_scatter(values_, n_, -1.0f, corr, nullptr, nullptr, nullptr,
+                      false, up.s_recv);
+      HIP_TRY(hipStreamSynchronize(up.s_recv));
+      zero_buf(values_, n_);
+      recv_snapshot(fd, corr);
+      hip_add_delta_scatter(up.delta, cfg_.delta_bf16, n_, values_, nullptr,
+                            nullptr, up.s_recv);
+      // corr += R directly from the (fp32-or-bf16) residual
+      hip_add_delta_scatter(up.delta, cfg_.delta_bf16, n_, corr, nullptr,
+                            nullptr, up.s_recv);
+      for (int i : {LK_LEFT, LK_RIGHT}) {
+        Link& c = links_[i];
+        if (c.state.load() == L_ACTIVE)
+          hip_add_scatter(corr, n_, 1.0f, nullptr, c.delta, nullptr, nullptr,
+                          cfg_.delta_bf16, up.s_recv);
+      }
+      HIP_TRY(hipStreamSynchronize(up.s_recv));
+      notify_all_dirty();
+    }
+  } else if (rejoin && (ah.flags & ACC_SNAPSHOT)) {
     // Reconciliation on rejoin: V := S + R, where S is the new parent's
     // snapshot and R = our unsent up-residual at the instant reconciliation
     // begins.  A concurrent user update u (add_from/fused_sgd) must land
@@ -653,7 +781,30 @@ void Engine::reconnect_loop() try {
   if (gpu()) destroy_link_graphs(up);
   up.abort.store(false);
   up.state.store(L_FREE);
-  drop_children();
+
+  // Subtree-preserving mode: keep live children attached through the
+  // rejoin instead of demoting them (they would otherwise re-join through
+  // the root one by one, each with a fresh snapshot).  Needs an extra
+  // fp32[n] scratch for the correction accumulator; if that cannot be
+  // had, fall back to the demote path.
+  rejoin_preserve_ = false;
+  if (cfg_.preserve_subtree && cfg_.snapshot_join &&
+      (links_[LK_LEFT].state.load() == L_ACTIVE ||
+       links_[LK_RIGHT].state.load() == L_ACTIVE)) {
+    if (gpu()) {
+      HIP_TRY(hipSetDevice(cfg_.device));
+      if (hipMalloc(&sub_corr_dev_, n_ * 4) == hipSuccess) {
+        rejoin_preserve_ = true;
+      } else {
+        (void)hipGetLastError();
+        sub_corr_dev_ = nullptr;
+      }
+    } else {
+      sub_corr_host_.assign(static_cast<size_t>(n_), 0.0f);
+      rejoin_preserve_ = true;
+    }
+  }
+  if (!rejoin_preserve_) drop_children();
 
   bool explicit_mode = !cfg_.explicit_parent.empty();
   sockaddr_in target = root_addr_;
@@ -680,9 +831,15 @@ void Engine::reconnect_loop() try {
           // the full inherited state.  slot := values (zero + add) under
           // the exclusive user-op lock — a plain add would double-count any
           // update that landed in the slot since drop_children zeroed it
-          // (ADVICE round 1, medium).
-          void* fwd0 = links_[LK_LEFT].provisioned ? links_[LK_LEFT].delta : nullptr;
-          void* fwd1 = links_[LK_RIGHT].provisioned ? links_[LK_RIGHT].delta : nullptr;
+          // (ADVICE round 1, medium).  Slots that stayed ACTIVE (subtree-
+          // preserving mode) keep their live residual untouched: this
+          // node's values never changed, so their deltas are still exact.
+          void* fwd0 = (links_[LK_LEFT].provisioned &&
+                        links_[LK_LEFT].state.load() != L_ACTIVE)
+                           ? links_[LK_LEFT].delta : nullptr;
+          void* fwd1 = (links_[LK_RIGHT].provisioned &&
+                        links_[LK_RIGHT].state.load() != L_ACTIVE)
+                           ? links_[LK_RIGHT].delta : nullptr;
           if (fwd0 || fwd1) {
             std::unique_lock<std::shared_mutex> ug(user_m_);
             if (gpu()) {
@@ -771,8 +928,16 @@ void Engine::reconnect_loop() try {
       hops = 0;
     }
   }
+  if (sub_corr_dev_) (void)hipFree(sub_corr_dev_), sub_corr_dev_ = nullptr;
+  sub_corr_host_.clear();
+  sub_corr_host_.shrink_to_fit();
+  rejoin_preserve_ = false;
   reconnecting_.store(false);
 } catch (const std::exception& e) {
+  if (sub_corr_dev_) (void)hipFree(sub_corr_dev_), sub_corr_dev_ = nullptr;
+  sub_corr_host_.clear();
+  sub_corr_host_.shrink_to_fit();
+  rejoin_preserve_ = false;
   set_error(std::string("reconnect failed: ") + e.what());
   reconnecting_.store(false);
 }
@@ -1163,7 +1328,7 @@ void Engine::send_snapshot(Link& lk) {
   }
 }
 
-void Engine::recv_snapshot(int fd) {
+void Engine::recv_snapshot(int fd, float* corr) {
   Link& up = links_[LK_UP];
   const int64_t chunk_bytes = gpu() ? (SA_ + P_) : (1 << 26);
   // even (prefer 64-aligned) chunk offsets: bf16 packed-pair atomics in the
@@ -1176,6 +1341,7 @@ void Engine::recv_snapshot(int fd) {
   void* fwd[2] = {
       links_[LK_LEFT].provisioned ? links_[LK_LEFT].delta : nullptr,
       links_[LK_RIGHT].provisioned ? links_[LK_RIGHT].delta : nullptr};
+  if (corr) fwd[0] = fwd[1] = nullptr;  // preserving: no raw-S forwarding
   for (int64_t off = 0; off < n_; off += chunk_elems) {
     int64_t ce = std::min(chunk_elems, n_ - off);
     if (gpu()) {
@@ -1190,6 +1356,10 @@ void Engine::recv_snapshot(int fd) {
                       values_ + off, fwd[0] ? doff(fwd[0], off) : nullptr,
                       fwd[1] ? doff(fwd[1], off) : nullptr, nullptr,
                       cfg_.delta_bf16, up.s_recv);
+      if (corr)  // correction accumulator is plain fp32
+        hip_add_scatter(reinterpret_cast<float*>(up.recv_buf), ce, 1.0f,
+                        corr + off, nullptr, nullptr, nullptr, false,
+                        up.s_recv);
       HIP_TRY(hipStreamSynchronize(up.s_recv));
     } else {
       if (!io_read(fd, tmp.data(), ce * 4))
@@ -1200,6 +1370,7 @@ void Engine::recv_snapshot(int fd) {
           float v = snap[i];
           if (v == 0.0f) continue;
           atomic_add_f32(values_ + off + i, v);
+          if (corr) corr[off + i] += v;  // corr is reconnect-thread-private
           if (fwd[0]) atomic_add_f32(fdelta(fwd[0]) + off + i, v);
           if (fwd[1]) atomic_add_f32(fdelta(fwd[1]) + off + i, v);
         }

@@ -127,6 +127,13 @@ class Engine {
   std::thread listen_thread_;
   std::thread reconnect_thread_;
   std::atomic<bool> reconnecting_{false};
+  // subtree-preserving rejoin (cfg.preserve_subtree): children stay
+  // attached; the reconciliation accumulates a correction delta
+  // corr = S + R - V_old (exact: concurrent updates never touch corr) that
+  // is added to every live child's residual and drains as ordinary gossip
+  bool rejoin_preserve_ = false;   // decided per rejoin attempt
+  float* sub_corr_dev_ = nullptr;  // GPU scratch fp32[n] (corr)
+  std::vector<float> sub_corr_host_;
   std::atomic<uint64_t> reconnects_{0};
   std::atomic<int> rccl_failures_{0};  // >=2: stop offering/requesting RCCL
   std::atomic<bool> closing_{false};
@@ -169,7 +176,11 @@ class Engine {
   void rccl_upgrade(Link& lk, const uint8_t* ids, bool is_parent);
   bool send_packet(Link& lk, const float* scales_host);  // post-quantize I/O
   void send_snapshot(Link& lk);
-  void recv_snapshot(int fd);
+  // corr == nullptr: snapshot chunks add into values + the child-slot
+  // residuals (join-state forwarding).  corr != nullptr (subtree-preserving
+  // rejoin): chunks add into values + corr ONLY — live children receive a
+  // single exact correction later instead of the raw snapshot.
+  void recv_snapshot(int fd, float* corr = nullptr);
   void apply_packet(Link& lk, const float* scales_host);
   void link_down(Link& lk, const std::string& why, bool remote);
   void set_error(const std::string& e);

@@ -37,7 +37,8 @@ class _SharedBase:
     def __init__(self, host: str, port: int, sizes: Sequence[int],
                  device: torch.device, codec: str = "1bit", *,
                  snapshot_join: bool = True, use_rccl: bool = True,
-                 reconnect: bool = False, keepalive_s: float = 1.0,
+                 reconnect: bool = False, preserve_subtree: bool = False,
+                 keepalive_s: float = 1.0,
                  bw_limit: float = 0.0, sync_interval_s: float = 0.0,
                  expected_children: int = 2,
                  provision_up: bool = True, explicit_parent: str = "",
@@ -57,6 +58,9 @@ class _SharedBase:
         cfg.snapshot_join = snapshot_join
         cfg.use_rccl = use_rccl
         cfg.reconnect = reconnect
+        if preserve_subtree and not snapshot_join:
+            raise ValueError("preserve_subtree requires snapshot_join")
+        cfg.preserve_subtree = preserve_subtree
         cfg.keepalive_s = keepalive_s
         cfg.bw_limit = float(bw_limit)
         cfg.min_round_interval_s = float(sync_interval_s)
