@@ -144,3 +144,45 @@ def test_child_carries_grandchild_through_master_restart():
                     p.kill()
         if m.is_alive():
             m.kill()
+
+
+def test_correction_algebra_is_exact():
+    """Pure simulation of the subtree-preserving reconciliation phases:
+    after the child drains its residual, its view equals the rejoiner's
+    values EXACTLY, for any V_old, S, R, U_c and any concurrent update u
+    landing during the lock-free snapshot window."""
+    torch.manual_seed(0)
+    n = 1000
+    V_old = torch.randn(n)   # rejoiner's pre-reconciliation replica
+    S = torch.randn(n)       # new parent's snapshot
+    R = torch.randn(n)       # unsent up-residual at capture
+    U_c = torch.randn(n)     # child-link residual (still exact vs V_old)
+    u = torch.randn(n)       # updates racing the snapshot window
+    child_view = V_old - U_c  # drained-invariant starting point
+
+    # phase 1 (exclusive): capture R, corr := -V_old, zero values/up.delta
+    tmpR = R.clone()
+    corr = -V_old.clone()
+    values = torch.zeros(n)
+    up_delta = torch.zeros(n)
+    child_delta = U_c.clone()  # untouched
+
+    # phase 2 (lock-free): snapshot adds into values+corr; u adds into
+    # values and every provisioned residual — never into corr
+    values += S
+    corr += S
+    values += u
+    up_delta += u
+    child_delta += u
+
+    # phase 3 (exclusive): re-add R; child gets the correction
+    values += tmpR
+    up_delta += tmpR
+    corr += tmpR
+    child_delta += corr
+
+    final_child = child_view + child_delta  # residual fully drained
+    torch.testing.assert_close(final_child, values, rtol=0, atol=1e-5)
+    torch.testing.assert_close(values, S + R + u, rtol=0, atol=1e-5)
+    # and the upward residual carries exactly R + u
+    torch.testing.assert_close(up_delta, R + u, rtol=0, atol=1e-5)
