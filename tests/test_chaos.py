@@ -31,11 +31,12 @@ def _fd_count():
         return -1
 
 
-def _chaos_rank(rank, port_base, stop_ev, q):
+def _chaos_rank(rank, port_base, stop_ev, q, preserve=False):
     try:
         sh = SharedFlat(
             "127.0.0.1", port_base, [N], device="cpu", codec="1bit",
             reconnect=True, snapshot_join=True, use_rccl=False,
+            preserve_subtree=preserve,
             expected_children=len(tree_children(rank, WORLD)),
             provision_up=rank > 0,
             explicit_parent=(f"127.0.0.1:{port_base + tree_parent(rank)}"
@@ -57,14 +58,20 @@ def _chaos_rank(rank, port_base, stop_ev, q):
         q.put(("fail", rank, repr(e), 0, 0, 0, ""))
 
 
-def test_tree_chaos_kill_restart_interior_ranks():
+import pytest
+
+
+@pytest.mark.parametrize("preserve", [False, True],
+                         ids=["demote", "preserve_subtree"])
+def test_tree_chaos_kill_restart_interior_ranks(preserve):
     port_base = free_port()
     ctx = mp.get_context("spawn")
     stop_ev = ctx.Event()
     q = ctx.Queue()
 
     def spawn(rank):
-        p = ctx.Process(target=_chaos_rank, args=(rank, port_base, stop_ev, q))
+        p = ctx.Process(target=_chaos_rank,
+                        args=(rank, port_base, stop_ev, q, preserve))
         p.start()
         return p
 
