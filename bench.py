@@ -341,6 +341,7 @@ def run_table(args, rank, world, device):
         f"{len(list(model.parameters()))} tensors")
     port_base = int(os.environ.get("SHTENS_PORT_BASE", 52000))
     nchild = len(tree_children(rank, world))
+    tj0 = time.perf_counter()
     sh = SharedTable(
         "127.0.0.1", port_base, {n: p.data for n, p in model.named_parameters()},
         codec=args.codec,
@@ -350,6 +351,7 @@ def run_table(args, rank, world, device):
         provision_up=rank > 0,
         explicit_parent=f"127.0.0.1:{port_base + tree_parent(rank)}" if rank else "",
         listen_port=port_base + rank if world > 1 else 0)
+    join_s = time.perf_counter() - tj0  # ctor includes join + snapshot
     # the table holds its own replica; free the model to fit 8B-scale runs
     # (2 x ~100 GB working set on one 288 GB device)
     del model
@@ -398,6 +400,8 @@ def run_table(args, rank, world, device):
             "n_tensors": len(sh.names), "codec": args.codec,
             "wire_gbps": round(wire / dt / 1e9, 3),
             "staleness_p50": s1["staleness_p50"],
+            # worst rank's table create incl. join + snapshot (master: alloc)
+            "join_s": round(max_over_ranks(dist, join_s if rank else 0.0), 2),
         },
     }
     sh.close()
