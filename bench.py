@@ -209,7 +209,7 @@ def run_train(args, rank, world, device):
     use_bf16_params = device.type == "cuda" and not args.fp32_params
     trainer = AsyncDPTrainer(
         model, host="127.0.0.1",
-        port_base=int(os.environ.get("SHTENS_PORT_BASE", 52000)),
+        port_base=int(os.environ.get("SHTENS_PORT_BASE", 21000)),
         rank=rank, world=world, lr=args.lr, momentum=0.9,
         optimizer=args.opt, weight_decay=0.01 if args.opt == "adamw" else 0.0,
         amp_dtype=torch.bfloat16 if device.type == "cuda" else None,
@@ -339,7 +339,7 @@ def run_table(args, rank, world, device):
     model = Llama(lcfg).to(device)
     log(f"llama table: {model.num_params()/1e6:.0f}M params, "
         f"{len(list(model.parameters()))} tensors")
-    port_base = int(os.environ.get("SHTENS_PORT_BASE", 52000))
+    port_base = int(os.environ.get("SHTENS_PORT_BASE", 21000))
     nchild = len(tree_children(rank, world))
     tj0 = time.perf_counter()
     sh = SharedTable(
@@ -419,7 +419,7 @@ def run_paramsync(args, rank, world, device):
     from sharedtensor_amd.parallel.async_dp import tree_children, tree_parent
 
     n = args.numel
-    port_base = int(os.environ.get("SHTENS_PORT_BASE", 52000))
+    port_base = int(os.environ.get("SHTENS_PORT_BASE", 21000))
     nchild = len(tree_children(rank, world))
     sh = SharedFlat(
         "127.0.0.1", port_base, [n], device=device, codec=args.codec,

@@ -67,7 +67,7 @@ def main():
     ap.add_argument("--child-threads", type=int, default=64)
     args = ap.parse_args()
     n = int(args.gb * 1e9 / 4)
-    port = 56231
+    port = 23231
     if os.path.exists("/tmp/shtens_cfg5_stop"):
         os.unlink("/tmp/shtens_cfg5_stop")
 

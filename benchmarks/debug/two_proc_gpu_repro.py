@@ -42,7 +42,7 @@ def child(port):
 
 def main():
     import sharedtensor_amd as st
-    port = 55881
+    port = 23881
     torch.cuda.set_device(0)
     log("master", "creating")
     master = st.create_or_fetch("127.0.0.1", port,

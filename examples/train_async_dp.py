@@ -62,7 +62,7 @@ def main():
     ap.add_argument("--lr", type=float, default=0.01)
     ap.add_argument("--opt", choices=["sgd", "adamw"], default="sgd")
     ap.add_argument("--codec", choices=["1bit", "fp8", "int4"], default="1bit")
-    ap.add_argument("--port", type=int, default=51000)
+    ap.add_argument("--port", type=int, default=22000)
     args = ap.parse_args()
     if "RANK" in os.environ:
         rank_main(args)

@@ -172,7 +172,7 @@ class AsyncDPTrainer:
         rank = int(os.environ.get("RANK", 0)) if rank is None else rank
         world = int(os.environ.get("WORLD_SIZE", 1)) if world is None else world
         if port_base is None:
-            port_base = int(os.environ.get("SHTENS_PORT_BASE", 52000))
+            port_base = int(os.environ.get("SHTENS_PORT_BASE", 21000))
         self.model = model
         if param_dtype is None:
             param_dtype = torch.float32

@@ -1,7 +1,6 @@
 """Async-DP trainer wiring on CPU with 2 processes (explicit tree topology —
 the same code path the multi-GPU bench uses, minus the GPU)."""
 import multiprocessing as mp
-import socket
 import time
 
 import torch
@@ -9,23 +8,9 @@ import torch
 from sharedtensor_amd.models.gpt2 import GPT2, GPT2Config
 from sharedtensor_amd.parallel.async_dp import (AsyncDPTrainer, tree_children,
                                                 tree_parent)
+from sharedtensor_amd.utils import free_port
 
 
-def free_port_base():
-    socks = []
-    base = None
-    for _ in range(20):
-        s = socket.socket()
-        s.bind(("127.0.0.1", 0))
-        p = s.getsockname()[1]
-        if p < 64000:
-            base = p
-            s.close()
-            break
-        socks.append(s)
-    for s in socks:
-        s.close()
-    return base
 
 
 def test_tree_helpers():
@@ -66,8 +51,7 @@ def _worker(rank, world, port_base, q):
 
 
 def test_two_rank_async_dp_cpu():
-    port_base = free_port_base()
-    assert port_base
+    port_base = free_port(span=2)
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
     procs = [ctx.Process(target=_worker, args=(r, 2, port_base, q))
@@ -88,8 +72,7 @@ def test_two_rank_async_dp_cpu():
 def test_four_rank_async_dp_cpu():
     """Depth-2 explicit tree (rank 0 <- 1,2; rank 1 <- 3): the same topology
     wiring the driver's 8-GPU scaling bench uses, exercised on CPU."""
-    port_base = free_port_base()
-    assert port_base
+    port_base = free_port(span=4)
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
     procs = [ctx.Process(target=_worker, args=(r, 4, port_base, q))

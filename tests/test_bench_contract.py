@@ -6,9 +6,12 @@ import subprocess
 import sys
 
 
-def run_bench(*extra, port_base="53611", timeout=240):
+def run_bench(*extra, port_base=None, timeout=240, span=1):
+    from sharedtensor_amd.utils import free_port
+    if port_base is None:
+        port_base = str(free_port(span=span))
     env = dict(os.environ)
-    env["SHTENS_PORT_BASE"] = port_base
+    env["SHTENS_PORT_BASE"] = str(port_base)
     env.pop("WORLD_SIZE", None)  # exercise the driver's plain invocation
     env.pop("RANK", None)
     out = subprocess.run(
@@ -40,7 +43,7 @@ def test_train_contract():
 
 
 def test_n1_paramsync_self_describing():
-    d = run_bench(port_base="53631")
+    d = run_bench()
     cfg = d["config"]
     assert cfg["paramsync"] == "n/a at n_gpus=1 (no links)"
     assert cfg["paramsync_wire_gbps"] is None
@@ -50,7 +53,7 @@ def test_n1_paramsync_self_describing():
 def test_self_launch_gpus2():
     """The driver's plain `python bench.py --gpus 2` must complete unaided:
     bench forks both ranks itself (no torchrun, no preset WORLD_SIZE)."""
-    d = run_bench("--gpus", "2", port_base="53641", timeout=420)
+    d = run_bench("--gpus", "2", timeout=420, span=2)
     assert d["n_gpus"] == 2
     assert d["config"]["parallelism"] == "async-dp2"
     # with links up, the param-sync half of the metric is measured
@@ -66,8 +69,7 @@ def test_self_launch_gpus2():
 def test_self_launch_gpus8_dry():
     """`--gpus 8` self-launch dry-run on CPU: the full 8-rank tree forms,
     steps, and reports without hanging (pre-stages the driver's SCALE run)."""
-    d = run_bench("--gpus", "8", "--steps", "1", port_base="53651",
-                  timeout=600)
+    d = run_bench("--gpus", "8", "--steps", "1", timeout=600, span=8)
     assert d["n_gpus"] == 8
     assert d["config"]["parallelism"] == "async-dp8"
 

@@ -64,7 +64,7 @@ import pytest
 @pytest.mark.parametrize("preserve", [False, True],
                          ids=["demote", "preserve_subtree"])
 def test_tree_chaos_kill_restart_interior_ranks(preserve):
-    port_base = free_port()
+    port_base = free_port(span=WORLD)
     ctx = mp.get_context("spawn")
     stop_ev = ctx.Event()
     q = ctx.Queue()

@@ -93,7 +93,7 @@ def _g_proc(port, q, go_ev, done_ev):
 
 def test_gpu_subtree_preserved_through_master_restart():
     torch.cuda.set_device(0)
-    port = free_port()
+    port = free_port(span=3)
     ctx = mp.get_context("spawn")
     ready_q = ctx.Queue()
     die_ev = ctx.Event()

@@ -88,7 +88,7 @@ def _g_proc(port, q, go_ev, done_ev):
 
 
 def test_child_carries_grandchild_through_master_restart():
-    port = free_port()
+    port = free_port(span=3)
     ctx = mp.get_context("spawn")
     ready_q = ctx.Queue()
     die_ev = ctx.Event()
